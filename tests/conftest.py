@@ -1,0 +1,44 @@
+import os
+import sys
+from pathlib import Path
+
+import pytest
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: requires a real MI355X GPU (run via gpurun)")
+
+
+@pytest.fixture
+def isolated_env(tmp_path, monkeypatch):
+    """Fully isolated clawker state dirs (reference: internal/testenv)."""
+    for var, sub in [
+        ("CLAWKER_CONFIG_DIR", "config"),
+        ("CLAWKER_DATA_DIR", "data"),
+        ("CLAWKER_STATE_DIR", "state"),
+        ("CLAWKER_RUNTIME_DIR", "run"),
+        ("CLAWKER_IMAGE_DIR", "images"),
+        ("CLAWKER_SANDBOX_DIR", "sandboxes"),
+        ("CLAWKER_VOLUME_DIR", "volumes"),
+    ]:
+        d = tmp_path / sub
+        d.mkdir(exist_ok=True)
+        monkeypatch.setenv(var, str(d))
+    return tmp_path
+
+
+def _isolation_available() -> bool:
+    """Can we create real sandboxes here (root + namespaces + overlayfs)?"""
+    if os.geteuid() != 0:
+        return False
+    import subprocess
+    r = subprocess.run(["unshare", "-pmf", "true"], capture_output=True)
+    return r.returncode == 0
+
+
+ISOLATION = _isolation_available()
+
+requires_isolation = pytest.mark.skipif(
+    not ISOLATION, reason="needs root + namespace support for real sandboxes")
