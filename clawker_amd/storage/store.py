@@ -199,7 +199,7 @@ def discover_project_layers(start: Path, stop_at: Path | None = None) -> list[Pa
     """
     from .. import consts
 
-    found: list[Path] = []
+    groups: list[list[Path]] = []
     cur = start.resolve()
     stop = stop_at.resolve() if stop_at else None
     while True:
@@ -207,22 +207,28 @@ def discover_project_layers(start: Path, stop_at: Path | None = None) -> list[Pa
         dot_form = cur / consts.PROJECT_FILE_NAME
         local_dir = cur / consts.PROJECT_DIR_NAME / consts.PROJECT_LOCAL_BASENAME
         local_dot = cur / (consts.PROJECT_LOCAL_BASENAME.replace("clawker", ".clawker", 1))
+        group: list[Path] = []
         if dir_form.is_file():
-            found.append(dir_form)
+            group.append(dir_form)
             if local_dir.is_file():
-                found.append(local_dir)
+                group.append(local_dir)
         elif dot_form.is_file():
-            found.append(dot_form)
+            group.append(dot_form)
             if local_dot.is_file():
-                found.append(local_dot)
+                group.append(local_dot)
+        if group:
+            groups.append(group)
         if stop is not None and cur == stop:
             break
         if cur.parent == cur:
             break
         cur = cur.parent
-    # farthest-from-start first so nearest (most specific) ends last / wins
-    found.reverse()
-    return found
+    # farthest-from-start dir first so nearest (most specific) ends last /
+    # wins; within a dir, the local layer stays above its clawker.yaml
+    out: list[Path] = []
+    for group in reversed(groups):
+        out.extend(group)
+    return out
 
 
 # ------------------------------------------------------------------ store ---
