@@ -1,0 +1,645 @@
+// ckd — PID 1 of every clawker-amd sandbox.
+//
+// The reference runs `clawkerd` as container PID 1 (clawkerd/ package:
+// mTLS gRPC Session stream, spawn with kernel-side privilege drop,
+// signal forwarding, two-phase zombie reaping — SURVEY.md §2.5). This
+// rebuild keeps those duties but speaks a framed-JSON protocol over a
+// Unix control socket in the sandbox's shared runtime dir (single-node
+// appliance: filesystem permissions replace the mTLS+OAuth stack; the
+// host side of the socket lives in a root-owned 0700 dir).
+//
+// Protocol (length-prefixed JSON frames; see clawker_amd/engine/wire.py):
+//   -> hello                         <- {t:hello, initialized, cmd_running}
+//   -> {t:agent_ready, cmd?:[..]}    spawn the agent CMD (once)
+//   -> {t:agent_initialized}         persist one-time init marker
+//   -> {t:exec, id, stages:[{argv,uid,gid,cwd}], stdin?, env?}
+//         <- {t:started,id} {t:out,id,stream,data(b64)}
+//            {t:stage_exit,id,idx,code} {t:done,id,code}
+//   -> {t:attach}                    subscribe to console stream
+//         <- {t:console, data(b64)}  (agent CMD output)
+//   -> {t:stdin, data(b64)} {t:close_stdin} {t:resize,rows,cols}
+//   -> {t:signal, sig}               signal the agent process group
+//   -> {t:status}                    <- {t:status, ...}
+//   <- {t:agent_exit, code}          broadcast when the agent CMD exits
+//
+// ckd exits with the agent's code (bash convention 128+sig) after a
+// bounded orphan drain; the pid-namespace teardown then kills stragglers.
+
+#include <errno.h>
+#include <fcntl.h>
+#include <grp.h>
+#include <poll.h>
+#include <pwd.h>
+#include <signal.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <sys/ioctl.h>
+#include <sys/stat.h>
+#include <sys/types.h>
+#include <sys/wait.h>
+#include <termios.h>
+#include <time.h>
+#include <unistd.h>
+
+#include <map>
+#include <string>
+#include <vector>
+
+#include "../common/minijson.hpp"
+#include "../common/util.hpp"
+
+using ck::die;
+using ck::warn;
+
+namespace {
+
+constexpr const char* kCtlSock = "/run/clawker/ctl.sock";
+constexpr const char* kConsoleLog = "/run/clawker/console.log";
+constexpr const char* kReadyDir = "/var/run/clawker";
+constexpr const char* kReadyFile = "/var/run/clawker/ready";
+constexpr const char* kInitMarkerDir = "/var/lib/clawker";
+constexpr const char* kInitMarker = "/var/lib/clawker/initialized";
+
+mj::Value g_spec;
+int g_selfpipe[2];    // SIGCHLD -> poll wakeup
+
+void sigchld(int) {
+  char b = 'c';
+  ssize_t r = write(g_selfpipe[1], &b, 1);
+  (void)r;
+}
+
+// ------------------------------------------------------------- users -------
+
+struct Cred { uid_t uid = 0; gid_t gid = 0; std::string home = "/"; std::string name = "root"; };
+
+Cred resolve_user(const std::string& spec) {
+  // "" or "root" => root; "uid:gid"; or /etc/passwd name
+  Cred c;
+  if (spec.empty() || spec == "root") return c;
+  size_t colon = spec.find(':');
+  if (colon != std::string::npos && spec.find_first_not_of("0123456789:") == std::string::npos) {
+    c.uid = atoi(spec.substr(0, colon).c_str());
+    c.gid = atoi(spec.substr(colon + 1).c_str());
+    c.name = spec;
+    return c;
+  }
+  if (struct passwd* pw = getpwnam(spec.c_str())) {
+    c.uid = pw->pw_uid;
+    c.gid = pw->pw_gid;
+    c.home = pw->pw_dir && *pw->pw_dir ? pw->pw_dir : "/";
+    c.name = spec;
+    return c;
+  }
+  warn("user %s not found; running as root", spec.c_str());
+  return c;
+}
+
+void drop_to(const Cred& c) {
+  // kernel-side drop ordering: groups -> gid -> uid (reference:
+  // clawkerd/spawn_unix.go SysProcAttr.Credential semantics)
+  if (c.uid == 0 && c.gid == 0) return;
+  if (initgroups(c.name.c_str(), c.gid) != 0) {
+    if (setgroups(0, nullptr) != 0) warn("setgroups");
+  }
+  if (setgid(c.gid) != 0) die("setgid %d", c.gid);
+  if (setuid(c.uid) != 0) die("setuid %d", c.uid);
+}
+
+std::vector<std::string> build_env(const mj::Value& env_obj, const Cred& c) {
+  std::vector<std::string> env;
+  bool has_path = false, has_home = false, has_user = false, has_term = false;
+  for (const auto& kv : env_obj.as_obj()) {
+    if (kv.first == "PATH") has_path = true;
+    if (kv.first == "HOME") has_home = true;
+    if (kv.first == "USER") has_user = true;
+    if (kv.first == "TERM") has_term = true;
+    env.push_back(kv.first + "=" + kv.second.as_str());
+  }
+  if (!has_path) env.push_back("PATH=/usr/local/sbin:/usr/local/bin:/usr/sbin:/usr/bin:/sbin:/bin");
+  if (!has_home) env.push_back("HOME=" + c.home);
+  if (!has_user) { env.push_back("USER=" + c.name); env.push_back("LOGNAME=" + c.name); }
+  if (!has_term) env.push_back("TERM=xterm-256color");
+  return env;
+}
+
+std::vector<char*> to_argv(const std::vector<std::string>& v) {
+  std::vector<char*> out;
+  for (const auto& s : v) out.push_back(const_cast<char*>(s.c_str()));
+  out.push_back(nullptr);
+  return out;
+}
+
+// ------------------------------------------------------------- clients -----
+
+struct Client {
+  int fd;
+  std::string buf;          // partial frame bytes
+  bool attached = false;    // subscribed to console stream
+};
+
+std::vector<Client> g_clients;
+
+void broadcast(const mj::Value& v, bool attached_only = false) {
+  for (auto& c : g_clients) {
+    if (attached_only && !c.attached) continue;
+    ck::send_frame(c.fd, v);
+  }
+}
+
+// --------------------------------------------------------------- agent -----
+
+struct Agent {
+  pid_t pid = -1;
+  int io = -1;              // pty master, or stdout pipe read end
+  int err = -1;             // stderr pipe (non-tty only)
+  int in = -1;              // stdin write end (non-tty; == io for tty)
+  bool tty = false;
+  bool running = false;
+  int exit_code = -1;
+  bool spawned = false;
+};
+
+Agent g_agent;
+int g_console_log = -1;
+
+void spawn_agent(const mj::Value& cmd_override) {
+  if (g_agent.spawned) return;
+  std::vector<std::string> argv;
+  const mj::Value& cmd = cmd_override.is_arr() && !cmd_override.as_arr().empty()
+                             ? cmd_override : g_spec["cmd"];
+  for (const auto& a : cmd.as_arr()) argv.push_back(a.as_str());
+  if (argv.empty()) argv = {"/bin/sh"};
+  Cred cred = resolve_user(g_spec["user"].as_str());
+  std::string workdir = g_spec["workdir"].as_str();
+  if (workdir.empty()) workdir = "/";
+  bool tty = g_spec["tty"].as_bool(false);
+  auto env = build_env(g_spec["env"], cred);
+
+  int master = -1, slave = -1;
+  int outp[2] = {-1, -1}, errp[2] = {-1, -1}, inp[2] = {-1, -1};
+  if (tty) {
+    master = posix_openpt(O_RDWR | O_NOCTTY | O_CLOEXEC);
+    if (master < 0) die("openpt");
+    grantpt(master);
+    unlockpt(master);
+    slave = open(ptsname(master), O_RDWR | O_NOCTTY);
+    if (slave < 0) die("pts open");
+  } else {
+    if (pipe2(outp, O_CLOEXEC) || pipe2(errp, O_CLOEXEC) || pipe2(inp, O_CLOEXEC))
+      die("pipe");
+  }
+
+  pid_t pid = fork();
+  if (pid < 0) die("fork");
+  if (pid == 0) {
+    setsid();
+    if (tty) {
+      ioctl(slave, TIOCSCTTY, 0);
+      dup2(slave, 0); dup2(slave, 1); dup2(slave, 2);
+      if (slave > 2) close(slave);
+      close(master);
+    } else {
+      dup2(inp[0], 0); dup2(outp[1], 1); dup2(errp[1], 2);
+    }
+    if (chdir(workdir.c_str()) != 0) {
+      // fall back to / rather than failing the spawn
+      if (chdir("/") != 0) _exit(111);
+    }
+    drop_to(cred);
+    auto eargv = to_argv(argv);
+    auto eenv = to_argv(env);
+    execvpe(eargv[0], eargv.data(), eenv.data());
+    fprintf(stderr, "ckd: exec %s: %s\n", eargv[0], strerror(errno));
+    _exit(127);
+  }
+  if (tty) {
+    close(slave);
+    fcntl(master, F_SETFL, O_NONBLOCK);
+    g_agent.io = master;
+    g_agent.in = master;
+  } else {
+    close(outp[1]); close(errp[1]); close(inp[0]);
+    fcntl(outp[0], F_SETFL, O_NONBLOCK);
+    fcntl(errp[0], F_SETFL, O_NONBLOCK);
+    g_agent.io = outp[0];
+    g_agent.err = errp[0];
+    g_agent.in = inp[1];
+  }
+  g_agent.pid = pid;
+  g_agent.tty = tty;
+  g_agent.running = true;
+  g_agent.spawned = true;
+}
+
+// ---------------------------------------------------------------- exec -----
+
+struct ExecJob {
+  std::string id;
+  int client_fd;
+  std::vector<pid_t> pids;
+  std::vector<int> codes;     // -1 = running
+  int out = -1, err = -1;     // last stage stdout, combined stderr
+  int pending_io = 0;
+};
+
+std::map<std::string, ExecJob> g_execs;
+
+void start_exec(Client& cl, const mj::Value& req) {
+  ExecJob job;
+  job.id = req["id"].as_str();
+  job.client_fd = cl.fd;
+  const auto& stages = req["stages"].as_arr();
+  if (stages.empty()) {
+    mj::Value e; e.set("t", "error").set("id", job.id).set("msg", "no stages");
+    ck::send_frame(cl.fd, e);
+    return;
+  }
+  std::string init_stdin;
+  if (req.has("stdin")) {
+    auto raw = ck::b64_decode(req["stdin"].as_str());
+    init_stdin.assign(raw.begin(), raw.end());
+  }
+
+  int in_fd = -1;   // read end feeding next stage's stdin
+  if (!init_stdin.empty()) {
+    int p[2];
+    if (pipe2(p, O_CLOEXEC)) die("pipe");
+    // write initial stdin from a detached writer child to avoid blocking
+    pid_t w = fork();
+    if (w == 0) {
+      close(p[0]);
+      ck::write_exact(p[1], init_stdin.data(), init_stdin.size());
+      _exit(0);
+    }
+    close(p[1]);
+    in_fd = p[0];
+  }
+
+  int errp[2];
+  if (pipe2(errp, O_CLOEXEC)) die("pipe");
+
+  for (size_t i = 0; i < stages.size(); i++) {
+    const mj::Value& st = stages[i];
+    bool last = i + 1 == stages.size();
+    int outp[2] = {-1, -1};
+    if (pipe2(outp, O_CLOEXEC)) die("pipe");
+
+    std::vector<std::string> argv;
+    for (const auto& a : st["argv"].as_arr()) argv.push_back(a.as_str());
+    Cred cred;
+    cred.uid = (uid_t)st["uid"].as_int(0);
+    cred.gid = (gid_t)st["gid"].as_int(0);
+    if (st.has("user") && !st["user"].as_str().empty())
+      cred = resolve_user(st["user"].as_str());
+    std::string cwd = st["cwd"].as_str();
+    auto env = build_env(req["env"], cred);
+
+    pid_t pid = fork();
+    if (pid < 0) die("fork");
+    if (pid == 0) {
+      if (in_fd >= 0) dup2(in_fd, 0);
+      else { int nul = open("/dev/null", O_RDONLY); dup2(nul, 0); }
+      dup2(outp[1], 1);
+      dup2(errp[1], 2);
+      if (!cwd.empty() && chdir(cwd.c_str()) != 0) _exit(126);
+      if (cred.uid || cred.gid) {
+        if (setgroups(0, nullptr) != 0) { /* best effort */ }
+        if (setgid(cred.gid) != 0 || setuid(cred.uid) != 0) _exit(126);
+      }
+      auto eargv = to_argv(argv);
+      auto eenv = to_argv(env);
+      execvpe(eargv[0], eargv.data(), eenv.data());
+      fprintf(stderr, "exec %s: %s\n", eargv[0], strerror(errno));
+      _exit(127);
+    }
+    job.pids.push_back(pid);
+    job.codes.push_back(-1);
+    if (in_fd >= 0) close(in_fd);
+    close(outp[1]);
+    if (last) {
+      fcntl(outp[0], F_SETFL, O_NONBLOCK);
+      job.out = outp[0];
+    } else {
+      in_fd = outp[0];   // next stage reads previous stdout
+    }
+  }
+  close(errp[1]);
+  fcntl(errp[0], F_SETFL, O_NONBLOCK);
+  job.err = errp[0];
+  job.pending_io = 2;
+
+  mj::Value started;
+  started.set("t", "started").set("id", job.id);
+  ck::send_frame(cl.fd, started);
+  g_execs[job.id] = job;
+}
+
+// drain an exec output fd; returns false when EOF
+bool pump_exec_fd(ExecJob& job, int which) {
+  int fd = which == 1 ? job.out : job.err;
+  if (fd < 0) return false;
+  char buf[65536];
+  while (true) {
+    ssize_t n = read(fd, buf, sizeof buf);
+    if (n > 0) {
+      mj::Value out;
+      out.set("t", "out").set("id", job.id).set("stream", (int64_t)which)
+         .set("data", ck::b64_encode(reinterpret_cast<uint8_t*>(buf), n));
+      ck::send_frame(job.client_fd, out);
+      continue;
+    }
+    if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return true;
+    // EOF or error
+    close(fd);
+    if (which == 1) job.out = -1; else job.err = -1;
+    job.pending_io--;
+    return false;
+  }
+}
+
+void maybe_finish_exec(ExecJob& job) {
+  if (job.pending_io > 0) return;
+  for (int c : job.codes)
+    if (c < 0) return;
+  mj::Value done;
+  done.set("t", "done").set("id", job.id).set("code", (int64_t)job.codes.back());
+  ck::send_frame(job.client_fd, done);
+  g_execs.erase(job.id);
+}
+
+// --------------------------------------------------------------- reaper ----
+
+void reap() {
+  // two-phase: identify the agent CMD specially, then drain any orphans
+  int wstatus;
+  pid_t pid;
+  while ((pid = waitpid(-1, &wstatus, WNOHANG)) > 0) {
+    int code = WIFEXITED(wstatus) ? WEXITSTATUS(wstatus)
+             : WIFSIGNALED(wstatus) ? 128 + WTERMSIG(wstatus) : 1;
+    if (pid == g_agent.pid) {
+      g_agent.running = false;
+      g_agent.exit_code = code;
+      continue;
+    }
+    for (auto& kv : g_execs) {
+      ExecJob& job = kv.second;
+      for (size_t i = 0; i < job.pids.size(); i++) {
+        if (job.pids[i] == pid && job.codes[i] < 0) {
+          job.codes[i] = code;
+          mj::Value se;
+          se.set("t", "stage_exit").set("id", job.id)
+            .set("idx", (int64_t)i).set("code", (int64_t)code);
+          ck::send_frame(job.client_fd, se);
+        }
+      }
+    }
+  }
+}
+
+// -------------------------------------------------------------- console ----
+
+void pump_console(int fd, int stream) {
+  char buf[65536];
+  while (true) {
+    ssize_t n = read(fd, buf, sizeof buf);
+    if (n > 0) {
+      if (g_console_log >= 0) ck::write_exact(g_console_log, buf, n);
+      mj::Value out;
+      out.set("t", "console").set("stream", (int64_t)stream)
+         .set("data", ck::b64_encode(reinterpret_cast<uint8_t*>(buf), n));
+      broadcast(out, /*attached_only=*/true);
+      continue;
+    }
+    if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return;
+    // EOF / EIO (pty closed)
+    if (stream == 1 && g_agent.io == fd) { close(fd); g_agent.io = -1; }
+    if (stream == 2 && g_agent.err == fd) { close(fd); g_agent.err = -1; }
+    return;
+  }
+}
+
+// ------------------------------------------------------------- handlers ----
+
+bool g_ready_announced = false;
+
+void handle_frame(Client& cl, const mj::Value& req) {
+  const std::string& t = req["t"].as_str();
+  if (t == "hello") {
+    mj::Value r;
+    r.set("t", "hello")
+     .set("initialized", ck::exists(kInitMarker))
+     .set("cmd_running", g_agent.running)
+     .set("pid", (int64_t)(g_agent.running ? g_agent.pid : -1))
+     .set("version", "0.1.0");
+    ck::send_frame(cl.fd, r);
+  } else if (t == "agent_ready") {
+    // CP releases the user CMD (reference: boot_steps.go AgentReady)
+    spawn_agent(req["cmd"]);
+    mj::Value r;
+    r.set("t", "ready_ack").set("pid", (int64_t)g_agent.pid);
+    ck::send_frame(cl.fd, r);
+  } else if (t == "agent_initialized") {
+    ck::mkdirs(kInitMarkerDir);
+    ck::write_file(kInitMarker, std::to_string(time(nullptr)));
+    mj::Value r; r.set("t", "ok");
+    ck::send_frame(cl.fd, r);
+  } else if (t == "exec") {
+    start_exec(cl, req);
+  } else if (t == "attach") {
+    cl.attached = true;
+    mj::Value r; r.set("t", "attached").set("tty", g_agent.tty);
+    ck::send_frame(cl.fd, r);
+  } else if (t == "stdin") {
+    auto data = ck::b64_decode(req["data"].as_str());
+    if (g_agent.in >= 0 && !data.empty())
+      ck::write_exact(g_agent.in, data.data(), data.size());
+  } else if (t == "close_stdin") {
+    if (!g_agent.tty && g_agent.in >= 0) { close(g_agent.in); g_agent.in = -1; }
+  } else if (t == "resize") {
+    if (g_agent.tty && g_agent.io >= 0) {
+      struct winsize ws{};
+      ws.ws_row = (unsigned short)req["rows"].as_int(24);
+      ws.ws_col = (unsigned short)req["cols"].as_int(80);
+      ioctl(g_agent.io, TIOCSWINSZ, &ws);
+    }
+  } else if (t == "signal") {
+    int sig = (int)req["sig"].as_int(SIGTERM);
+    if (g_agent.pid > 0) kill(-g_agent.pid, sig);   // whole process group
+  } else if (t == "status") {
+    mj::Value r;
+    r.set("t", "status").set("cmd_running", g_agent.running)
+     .set("exit_code", (int64_t)g_agent.exit_code)
+     .set("initialized", ck::exists(kInitMarker));
+    ck::send_frame(cl.fd, r);
+  } else {
+    mj::Value r; r.set("t", "error").set("msg", "unknown command: " + t);
+    ck::send_frame(cl.fd, r);
+  }
+}
+
+void drain_client(Client& cl, bool& drop) {
+  char buf[65536];
+  while (true) {
+    ssize_t n = read(cl.fd, buf, sizeof buf);
+    if (n > 0) {
+      cl.buf.append(buf, n);
+      // extract complete frames
+      while (cl.buf.size() >= 4) {
+        size_t len = (size_t(uint8_t(cl.buf[0])) << 24) | (size_t(uint8_t(cl.buf[1])) << 16) |
+                     (size_t(uint8_t(cl.buf[2])) << 8) | uint8_t(cl.buf[3]);
+        if (len > ck::kMaxFrame) { drop = true; return; }
+        if (cl.buf.size() < 4 + len) break;
+        std::string body = cl.buf.substr(4, len);
+        cl.buf.erase(0, 4 + len);
+        try {
+          handle_frame(cl, mj::parse(body));
+        } catch (const std::exception& e) {
+          mj::Value r; r.set("t", "error").set("msg", e.what());
+          ck::send_frame(cl.fd, r);
+        }
+      }
+      continue;
+    }
+    if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return;
+    drop = true;   // EOF or error
+    return;
+  }
+}
+
+}  // namespace
+
+int main() {
+  const char* spec_path = getenv("CKD_SPEC");
+  if (!spec_path) die("CKD_SPEC not set");
+  g_spec = mj::parse(ck::read_file(spec_path));
+
+  if (pipe2(g_selfpipe, O_CLOEXEC | O_NONBLOCK) != 0) die("selfpipe");
+  struct sigaction sa{};
+  sa.sa_handler = sigchld;
+  sa.sa_flags = SA_RESTART | SA_NOCLDSTOP;
+  sigaction(SIGCHLD, &sa, nullptr);
+  signal(SIGPIPE, SIG_IGN);
+
+  // forward termination signals to the agent's process group (reference:
+  // clawkerd signal forwarding excl. SIGCHLD/SIGURG)
+  auto fwd = [](int sig) {
+    if (g_agent.pid > 0) kill(-g_agent.pid, sig);
+    else if (sig == SIGTERM || sig == SIGINT) _exit(128 + sig);
+  };
+  struct sigaction fsa{};
+  fsa.sa_handler = fwd;
+  fsa.sa_flags = SA_RESTART;
+  for (int sig : {SIGTERM, SIGINT, SIGHUP, SIGQUIT, SIGUSR1, SIGUSR2})
+    sigaction(sig, &fsa, nullptr);
+
+  int listen_fd = ck::unix_listen(kCtlSock);
+  if (listen_fd < 0) die("listen %s", kCtlSock);
+  fcntl(listen_fd, F_SETFL, O_NONBLOCK);
+
+  g_console_log = open(kConsoleLog, O_WRONLY | O_CREAT | O_APPEND | O_CLOEXEC, 0600);
+
+  // ready file: the HEALTHCHECK analog (reference: Dockerfile.base.tmpl:245)
+  ck::mkdirs(kReadyDir);
+  ck::write_file(kReadyFile, "1");
+
+  if (g_spec["autostart"].as_bool(false)) spawn_agent(mj::Value());
+
+  bool exiting = false;
+  int64_t exit_deadline_ms = 0;
+  auto now_ms = [] {
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return (int64_t)ts.tv_sec * 1000 + ts.tv_nsec / 1000000;
+  };
+
+  while (true) {
+    std::vector<pollfd> pfds;
+    pfds.push_back({g_selfpipe[0], POLLIN, 0});
+    pfds.push_back({listen_fd, POLLIN, 0});
+    size_t client_base = pfds.size();
+    for (auto& c : g_clients) pfds.push_back({c.fd, POLLIN, 0});
+    int agent_io_idx = -1, agent_err_idx = -1;
+    if (g_agent.io >= 0) { agent_io_idx = pfds.size(); pfds.push_back({g_agent.io, POLLIN, 0}); }
+    if (g_agent.err >= 0) { agent_err_idx = pfds.size(); pfds.push_back({g_agent.err, POLLIN, 0}); }
+    std::vector<std::pair<std::string, int>> exec_fds;   // (id, which)
+    for (auto& kv : g_execs) {
+      if (kv.second.out >= 0) { exec_fds.push_back({kv.first, 1}); pfds.push_back({kv.second.out, POLLIN, 0}); }
+      if (kv.second.err >= 0) { exec_fds.push_back({kv.first, 2}); pfds.push_back({kv.second.err, POLLIN, 0}); }
+    }
+
+    int timeout = exiting ? 50 : 1000;
+    int rc = poll(pfds.data(), pfds.size(), timeout);
+    if (rc < 0 && errno != EINTR) die("poll");
+
+    // SIGCHLD wakeup
+    if (pfds[0].revents & POLLIN) {
+      char b[64];
+      while (read(g_selfpipe[0], b, sizeof b) > 0) {}
+    }
+    reap();
+
+    // accept new control clients
+    if (pfds[1].revents & POLLIN) {
+      int cfd;
+      while ((cfd = accept4(listen_fd, nullptr, nullptr, SOCK_NONBLOCK | SOCK_CLOEXEC)) >= 0)
+        g_clients.push_back(Client{cfd});
+    }
+
+    // client traffic
+    std::vector<int> drop_fds;
+    for (size_t i = 0; i < g_clients.size(); i++) {
+      if (client_base + i >= pfds.size()) break;
+      auto& pe = pfds[client_base + i];
+      if (pe.revents & (POLLIN | POLLHUP | POLLERR)) {
+        bool drop = false;
+        drain_client(g_clients[i], drop);
+        if (drop) drop_fds.push_back(g_clients[i].fd);
+      }
+    }
+    for (int fd : drop_fds) {
+      for (size_t i = 0; i < g_clients.size(); i++) {
+        if (g_clients[i].fd == fd) {
+          close(fd);
+          g_clients.erase(g_clients.begin() + i);
+          break;
+        }
+      }
+      // orphan exec jobs owned by a gone client keep running; their
+      // output is discarded when send fails (SIGPIPE ignored)
+    }
+
+    // console output
+    if (agent_io_idx >= 0 && (pfds[agent_io_idx].revents & (POLLIN | POLLHUP)))
+      pump_console(g_agent.io, 1);
+    if (agent_err_idx >= 0 && (pfds[agent_err_idx].revents & (POLLIN | POLLHUP)))
+      pump_console(g_agent.err, 2);
+
+    // exec output
+    for (auto& [id, which] : exec_fds) {
+      auto it = g_execs.find(id);
+      if (it != g_execs.end()) pump_exec_fd(it->second, which);
+    }
+    // exec completion
+    for (auto it = g_execs.begin(); it != g_execs.end();) {
+      auto cur = it++;
+      maybe_finish_exec(cur->second);
+    }
+
+    // agent exit: announce once, then drain orphans briefly and leave —
+    // pidns teardown reclaims anything still running
+    if (g_agent.spawned && !g_agent.running && !exiting) {
+      // flush any trailing console output first
+      if (g_agent.io >= 0) pump_console(g_agent.io, 1);
+      if (g_agent.err >= 0) pump_console(g_agent.err, 2);
+      mj::Value ev;
+      ev.set("t", "agent_exit").set("code", (int64_t)g_agent.exit_code);
+      broadcast(ev);
+      exiting = true;
+      exit_deadline_ms = now_ms() + 500;
+    }
+    if (exiting && now_ms() >= exit_deadline_ms)
+      return g_agent.exit_code < 0 ? 0 : g_agent.exit_code;
+  }
+}
