@@ -1,0 +1,416 @@
+"""The sandbox engine: create/start/stop/remove/exec/logs over the native
+runtime (ckrt/ckd).
+
+Reference analog: pkg/whail (Docker SDK wrapper with managed-label
+isolation, engine.go:40) + internal/docker middleware. Here there is no
+dockerd: the engine drives ckrt directly, so "managed isolation" is
+structural — the engine only ever sees sandboxes in its own state DB and
+store directories.
+"""
+from __future__ import annotations
+
+import os
+import shutil
+import signal
+import subprocess
+import time
+from dataclasses import dataclass
+from pathlib import Path
+
+from .. import consts
+from ..errors import ConflictError, EngineError, NotFoundError
+from ..logger import get as get_logger
+from .ckd_client import CkdClient
+from .images import HOSTFS, ImageStore
+from .spec import Device, Mount, SandboxSpec
+from .state import StateDB
+
+log = get_logger("engine")
+
+# virtual filesystems never passed through into sandboxes
+_VIRTUAL_FSTYPES = {
+    "proc", "sysfs", "devtmpfs", "devpts", "tmpfs", "cgroup", "cgroup2",
+    "mqueue", "debugfs", "tracefs", "securityfs", "pstore", "bpf",
+    "hugetlbfs", "configfs", "fusectl", "binfmt_misc", "autofs", "overlay",
+    "squashfs", "ramfs", "rpc_pipefs", "nsfs", "efivarfs",
+}
+_EXCLUDED_PREFIXES = ("/proc", "/sys", "/dev", "/run", "/tmp", "/boot",
+                      "/var/lib/docker", "/var/run")
+
+
+def native_bin_dir() -> Path:
+    v = os.environ.get("CLAWKER_NATIVE_BIN")
+    if v:
+        return Path(v)
+    return Path(__file__).resolve().parents[2] / "native" / "bin"
+
+
+def host_passthrough_mounts() -> list[Mount]:
+    """Separate real filesystems mounted under / that the hostfs overlay
+    lowerdir will NOT traverse (overlayfs ignores lowerdir submounts):
+    rebind them read-only so e.g. a separately-mounted /opt/rocm stays
+    visible inside sandboxes."""
+    out: list[Mount] = []
+    seen: set[str] = set()
+    try:
+        lines = Path("/proc/self/mountinfo").read_text().splitlines()
+    except OSError:
+        return out
+    for line in lines:
+        # mountinfo: ... mountpoint ... - fstype source opts
+        try:
+            left, right = line.split(" - ", 1)
+            fields = left.split()
+            mountpoint = fields[4]
+            fstype = right.split()[0]
+        except (ValueError, IndexError):
+            continue
+        if mountpoint == "/" or fstype in _VIRTUAL_FSTYPES:
+            continue
+        if any(mountpoint == p or mountpoint.startswith(p + "/") for p in _EXCLUDED_PREFIXES):
+            continue
+        if any(mountpoint.startswith(s + "/") or mountpoint == s for s in seen):
+            continue   # covered by an already-included ancestor (rbind)
+        seen.add(mountpoint)
+        out.append(Mount(src=mountpoint, dst=mountpoint, ro=True))
+    return out
+
+
+@dataclass
+class SandboxInfo:
+    name: str
+    project: str
+    agent: str
+    image: str
+    created: float
+    state: str          # created | running | exited | dead
+    exit_code: int | None
+    pid: int | None
+    gpus: list[int]
+    labels: dict
+    rundir: Path
+    statedir: Path
+
+
+class Engine:
+    def __init__(self, images: ImageStore | None = None, db: StateDB | None = None):
+        self.images = images or ImageStore()
+        self.db = db or StateDB()
+        self.rt_root = consts.runtime_dir() / "sandboxes"
+        self.rt_root.mkdir(parents=True, exist_ok=True)
+        self.store_root = consts.sandbox_store_dir()
+        self.store_root.mkdir(parents=True, exist_ok=True)
+        self._passthrough: list[Mount] | None = None
+
+    def close(self) -> None:
+        self.db.close()
+
+    # ------------------------------------------------------------- paths ----
+    def rundir(self, name: str) -> Path:
+        return self.rt_root / name
+
+    def statedir(self, name: str) -> Path:
+        return self.store_root / name
+
+    def ctl_sock(self, name: str) -> Path:
+        return self.rundir(name) / consts.CKD_SOCK_NAME
+
+    # ------------------------------------------------------------ create ----
+    def create(self, spec: SandboxSpec, image: str = HOSTFS) -> SandboxInfo:
+        if not spec.name.startswith(consts.SANDBOX_NAME_PREFIX):
+            raise EngineError("create", f"unmanaged name: {spec.name}")
+        if self.db.get_sandbox(spec.name) is not None:
+            raise ConflictError(f"sandbox exists: {spec.name}")
+        meta = self.images.get(image)
+
+        rundir = self.rundir(spec.name)
+        statedir = self.statedir(spec.name)
+        for d in (rundir / "bin", statedir / "upper", statedir / "work"):
+            d.mkdir(parents=True, exist_ok=True)
+        os.chmod(rundir, 0o700)
+
+        # stage the PID-1 supervisor into the rundir (bind-mounted at
+        # /run/clawker inside; ckrt execs /run/clawker/bin/ckd)
+        ckd_src = native_bin_dir() / "ckd"
+        if not ckd_src.is_file():
+            raise EngineError("create", f"ckd binary missing: {ckd_src} (run `make native`)")
+        ckd_dst = rundir / "bin" / "ckd"
+        if not ckd_dst.exists() or ckd_dst.stat().st_mtime < ckd_src.stat().st_mtime:
+            shutil.copy2(ckd_src, ckd_dst)
+
+        # per-sandbox identity files
+        (rundir / "hostname").write_text(spec.hostname + "\n")
+        (rundir / "hosts").write_text(
+            f"127.0.0.1\tlocalhost {spec.hostname}\n::1\tlocalhost\n")
+        if not (rundir / "resolv.conf").exists():
+            # default: loopback stub (the firewall dnsd path fills this in);
+            # without firewall the sandbox shares the host netns and this
+            # file is not written at all (host resolv.conf shows through)
+            if spec.netns:
+                (rundir / "resolv.conf").write_text("nameserver 127.0.0.1\n")
+
+        # rootfs stack from the image + host passthrough binds
+        spec.rundir = str(rundir)
+        spec.lowerdirs = self.images.lowerdirs_for(image)
+        spec.upper = str(statedir / "upper")
+        spec.work = str(statedir / "work")
+        spec.merged = str(rundir / "merged")
+        if self._passthrough is None:
+            self._passthrough = host_passthrough_mounts()
+        spec.mounts = self._passthrough + spec.mounts
+
+        # image-level env/user/cmd defaults
+        env = dict(meta.env)
+        env.update(spec.env)
+        spec.env = env
+        if not spec.user:
+            spec.user = meta.user
+        if not spec.cmd:
+            spec.cmd = list(meta.cmd)
+        if spec.workdir in ("", "/") and meta.workdir:
+            spec.workdir = meta.workdir
+
+        spec.labels.setdefault(consts.MANAGED_LABEL, "true")
+        spec_path = rundir / "spec.json"
+        spec.write(spec_path)
+
+        self.db.add_sandbox(
+            spec.name, spec.labels.get(consts.PROJECT_LABEL, ""),
+            spec.labels.get(consts.AGENT_LABEL, ""), image, spec.labels,
+            [int(x) for x in spec.labels.get(consts.GPU_LABEL, "").split(",") if x != ""],
+            str(spec_path), str(rundir), str(statedir))
+        log.info("sandbox_created", sandbox=spec.name, image=image)
+        return self.inspect(spec.name)
+
+    # ------------------------------------------------------------- start ----
+    def start(self, name: str, wait_ready_s: float = 10.0) -> SandboxInfo:
+        row = self._row(name)
+        rundir = Path(row["rundir"])
+        status = self._status(rundir)
+        if status.get("state") == "running" and self._pid_alive(status.get("pid")):
+            raise ConflictError(f"sandbox already running: {name}")
+        # clear stale run state
+        for f in ("exit.json", "status.json", "pid", "console.log"):
+            (rundir / f).unlink(missing_ok=True)
+
+        ckrt = native_bin_dir() / "ckrt"
+        if not ckrt.is_file():
+            raise EngineError("start", f"ckrt binary missing: {ckrt} (run `make native`)")
+        shim_log = open(rundir / "shim.log", "ab")
+        proc = subprocess.Popen(
+            [str(ckrt), "run", row["spec_path"]],
+            stdin=subprocess.DEVNULL, stdout=shim_log, stderr=shim_log,
+            start_new_session=True, close_fds=True)
+        shim_log.close()
+
+        # wait until ckd binds its control socket (or the sandbox dies fast)
+        sock = rundir / consts.CKD_SOCK_NAME
+        deadline = time.monotonic() + wait_ready_s
+        while time.monotonic() < deadline:
+            if sock.exists():
+                break
+            if (rundir / "exit.json").exists() or proc.poll() is not None:
+                tail = self._tail(rundir / "shim.log")
+                raise EngineError("start", f"sandbox died during boot: {tail}")
+            time.sleep(0.002)
+        else:
+            raise EngineError("start", f"timed out waiting for ckd socket ({name})")
+        log.info("sandbox_started", sandbox=name, shim_pid=proc.pid)
+        return self.inspect(name)
+
+    # -------------------------------------------------------------- stop ----
+    def stop(self, name: str, timeout_s: float = 10.0) -> int | None:
+        row = self._row(name)
+        rundir = Path(row["rundir"])
+        pid = self._init_pid(rundir)
+        if pid is None or not self._pid_alive(pid):
+            return self._exit_code(rundir)
+        try:
+            os.kill(pid, signal.SIGTERM)
+        except ProcessLookupError:
+            return self._exit_code(rundir)
+        deadline = time.monotonic() + timeout_s
+        while time.monotonic() < deadline:
+            if not self._pid_alive(pid):
+                return self._exit_code(rundir)
+            time.sleep(0.02)
+        try:
+            os.kill(pid, signal.SIGKILL)
+        except ProcessLookupError:
+            pass
+        self._wait_pid_gone(pid, 5.0)
+        log.info("sandbox_stopped", sandbox=name, forced=True)
+        return self._exit_code(rundir)
+
+    def kill(self, name: str, sig: int = signal.SIGKILL) -> None:
+        pid = self._init_pid(self.rundir(name))
+        if pid is not None and self._pid_alive(pid):
+            os.kill(pid, sig)
+
+    # -------------------------------------------------------------- wait ----
+    def wait(self, name: str, timeout_s: float | None = None) -> int:
+        row = self._row(name)
+        rundir = Path(row["rundir"])
+        deadline = None if timeout_s is None else time.monotonic() + timeout_s
+        while True:
+            code = self._exit_code(rundir)
+            if code is not None:
+                return code
+            pid = self._init_pid(rundir)
+            if pid is not None and not self._pid_alive(pid):
+                # shim writes exit.json right after reaping; brief grace
+                for _ in range(100):
+                    code = self._exit_code(rundir)
+                    if code is not None:
+                        return code
+                    time.sleep(0.01)
+                return -1
+            if deadline is not None and time.monotonic() > deadline:
+                raise EngineError("wait", f"timeout waiting for {name}")
+            time.sleep(0.02)
+
+    # ------------------------------------------------------------ remove ----
+    def remove(self, name: str, force: bool = False) -> None:
+        row = self._row(name)
+        rundir = Path(row["rundir"])
+        pid = self._init_pid(rundir)
+        if pid is not None and self._pid_alive(pid):
+            if not force:
+                raise ConflictError(f"sandbox running: {name} (use --force)")
+            self.stop(name, timeout_s=3.0)
+        shutil.rmtree(rundir, ignore_errors=True)
+        shutil.rmtree(row["statedir"], ignore_errors=True)
+        self.db.remove_sandbox(name)
+        log.info("sandbox_removed", sandbox=name)
+
+    # ----------------------------------------------------------- inspect ----
+    def inspect(self, name: str) -> SandboxInfo:
+        row = self._row(name)
+        rundir = Path(row["rundir"])
+        status = self._status(rundir)
+        state = "created"
+        pid = None
+        exit_code = self._exit_code(rundir)
+        if status.get("state") == "running":
+            pid = int(status["pid"])
+            if self._pid_alive(pid):
+                state = "running"
+            else:
+                state, pid = ("exited", None) if exit_code is not None else ("dead", None)
+        elif exit_code is not None:
+            state = "exited"
+        return SandboxInfo(
+            name=name, project=row["project"], agent=row["agent"], image=row["image"],
+            created=row["created"], state=state, exit_code=exit_code, pid=pid,
+            gpus=row["gpus"], labels=row["labels"], rundir=rundir,
+            statedir=Path(row["statedir"]))
+
+    def list(self, project: str | None = None, all_states: bool = True,
+             label_filters: dict | None = None) -> list[SandboxInfo]:
+        out = []
+        for row in self.db.list_sandboxes(project=project, label_filters=label_filters):
+            info = self.inspect(row["name"])
+            if not all_states and info.state != "running":
+                continue
+            out.append(info)
+        return out
+
+    # ------------------------------------------------------------- logs  ----
+    def logs(self, name: str) -> bytes:
+        row = self._row(name)
+        p = Path(row["rundir"]) / "console.log"
+        return p.read_bytes() if p.exists() else b""
+
+    # ------------------------------------------------------------ client ----
+    def client(self, name: str, deadline_s: float = 10.0,
+               timeout: float | None = 30.0) -> CkdClient:
+        self._row(name)
+        return CkdClient.wait_connect(self.ctl_sock(name), deadline_s, timeout=timeout)
+
+    def exec(self, name: str, argv: list[str], user: str = "",
+             stdin: bytes = b"", cwd: str = "") -> tuple[int, bytes, bytes]:
+        with self.client(name) as c:
+            stage: dict = {"argv": argv}
+            if user:
+                stage["user"] = user
+            if cwd:
+                stage["cwd"] = cwd
+            return c.exec([stage], stdin=stdin)
+
+    # ------------------------------------------------------------ volumes ---
+    def ensure_volume(self, name: str, labels: dict | None = None) -> tuple[Path, bool]:
+        """Create-or-get a named volume; returns (path, fresh)."""
+        row = self.db.get_volume(name)
+        if row is not None:
+            return Path(row["path"]), False
+        path = consts.volume_store_dir() / name
+        path.mkdir(parents=True, exist_ok=True)
+        self.db.add_volume(name, str(path), labels or {consts.MANAGED_LABEL: "true"})
+        return path, True
+
+    def remove_volume(self, name: str) -> None:
+        row = self.db.get_volume(name)
+        if row is None:
+            raise NotFoundError(f"volume not found: {name}")
+        shutil.rmtree(row["path"], ignore_errors=True)
+        self.db.remove_volume(name)
+
+    # ------------------------------------------------------------ helpers ---
+    def _row(self, name: str) -> dict:
+        row = self.db.get_sandbox(name)
+        if row is None:
+            raise NotFoundError(f"sandbox not found: {name}")
+        return row
+
+    @staticmethod
+    def _status(rundir: Path) -> dict:
+        p = rundir / "status.json"
+        try:
+            import json
+            return json.loads(p.read_text())
+        except (OSError, ValueError):
+            return {}
+
+    @staticmethod
+    def _exit_code(rundir: Path) -> int | None:
+        p = rundir / "exit.json"
+        try:
+            import json
+            return int(json.loads(p.read_text())["code"])
+        except (OSError, ValueError, KeyError):
+            return None
+
+    @staticmethod
+    def _init_pid(rundir: Path) -> int | None:
+        try:
+            return int((rundir / "pid").read_text().strip())
+        except (OSError, ValueError):
+            return None
+
+    @staticmethod
+    def _pid_alive(pid: int | None) -> bool:
+        if not pid:
+            return False
+        try:
+            os.kill(int(pid), 0)
+            return True
+        except (ProcessLookupError, PermissionError):
+            return False
+
+    @staticmethod
+    def _wait_pid_gone(pid: int, timeout_s: float) -> None:
+        deadline = time.monotonic() + timeout_s
+        while time.monotonic() < deadline:
+            try:
+                os.kill(pid, 0)
+            except ProcessLookupError:
+                return
+            time.sleep(0.01)
+
+    @staticmethod
+    def _tail(path: Path, n: int = 400) -> str:
+        try:
+            data = path.read_bytes()
+            return data[-n:].decode(errors="replace")
+        except OSError:
+            return ""

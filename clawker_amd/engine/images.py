@@ -1,0 +1,168 @@
+"""Content-addressed overlay image store.
+
+The reference builds Docker images (internal/docker/builder.go + bundler
+Dockerfile generation). This node has no Docker and no network, so images
+are *overlayfs layer stacks over the host filesystem* ("hostfs base"): the
+ROCm dev userland already on the node is the base; a build runs its steps
+in a throwaway sandbox whose upper dir becomes a new content-addressed
+layer; instantiating a sandbox is just an overlay mount of
+hostfs + image layers (ro) + a per-sandbox upper (rw). Zero pull, zero
+copy on create — this is where the cold-start win comes from.
+
+Store layout:
+  images/layers/<id>/fs/      layer content
+  images/meta/<name>.json     {"name","layers":[bottom..top],"env","user",
+                               "cmd","workdir","labels","created","parent"}
+"""
+from __future__ import annotations
+
+import hashlib
+import json
+import os
+import shutil
+import time
+import uuid
+from dataclasses import dataclass, field
+from pathlib import Path
+
+from .. import consts
+from ..errors import ConflictError, NotFoundError
+
+HOSTFS = "hostfs"
+
+
+def _enc(name: str) -> str:
+    return name.replace("/", "%2F").replace(":", "%3A")
+
+
+@dataclass
+class ImageMeta:
+    name: str
+    layers: list[str] = field(default_factory=list)   # bottom..top layer ids
+    env: dict = field(default_factory=dict)
+    user: str = ""
+    cmd: list = field(default_factory=list)
+    workdir: str = ""
+    labels: dict = field(default_factory=dict)
+    created: float = 0.0
+    parent: str = HOSTFS
+    base_hash: str = ""     # staleness check input hash (reference: basehash.go)
+
+    def to_dict(self) -> dict:
+        return self.__dict__.copy()
+
+    @classmethod
+    def from_dict(cls, d: dict) -> "ImageMeta":
+        known = {f for f in cls.__dataclass_fields__}  # type: ignore[attr-defined]
+        return cls(**{k: v for k, v in d.items() if k in known})
+
+
+class ImageStore:
+    def __init__(self, root: Path | None = None):
+        self.root = root or consts.image_store_dir()
+        (self.root / "layers").mkdir(parents=True, exist_ok=True)
+        (self.root / "meta").mkdir(parents=True, exist_ok=True)
+
+    # -- metadata ------------------------------------------------------------
+    def _meta_path(self, name: str) -> Path:
+        return self.root / "meta" / (_enc(name) + ".json")
+
+    def exists(self, name: str) -> bool:
+        return name == HOSTFS or self._meta_path(name).is_file()
+
+    def get(self, name: str) -> ImageMeta:
+        if name == HOSTFS:
+            return ImageMeta(name=HOSTFS, created=0.0, parent="")
+        p = self._meta_path(name)
+        if not p.is_file():
+            raise NotFoundError(f"image not found: {name}")
+        return ImageMeta.from_dict(json.loads(p.read_text()))
+
+    def put(self, meta: ImageMeta, overwrite: bool = True) -> None:
+        meta.created = meta.created or time.time()
+        p = self._meta_path(meta.name)
+        if p.exists() and not overwrite:
+            raise ConflictError(f"image exists: {meta.name}")
+        p.write_text(json.dumps(meta.to_dict(), indent=1))
+
+    def list(self) -> list[ImageMeta]:
+        out = []
+        for p in sorted((self.root / "meta").glob("*.json")):
+            out.append(ImageMeta.from_dict(json.loads(p.read_text())))
+        return out
+
+    def remove(self, name: str, prune_layers: bool = True) -> None:
+        meta = self.get(name)
+        self._meta_path(name).unlink(missing_ok=True)
+        if prune_layers:
+            self.prune_layers()
+
+    def tag(self, src: str, dst: str) -> None:
+        meta = self.get(src)
+        meta.name = dst
+        self.put(meta)
+
+    # -- layers ----------------------------------------------------------------
+    def new_layer_dir(self) -> tuple[str, Path]:
+        """Allocate a staging layer; caller fills fs/ then commit_layer()."""
+        lid = "tmp-" + uuid.uuid4().hex[:12]
+        d = self.root / "layers" / lid / "fs"
+        d.mkdir(parents=True)
+        return lid, d
+
+    def commit_layer(self, tmp_id: str) -> str:
+        """Rename a staging layer to its content id (cheap pseudo-hash:
+        file list + sizes + mtimes digest — enough for staleness checks)."""
+        src = self.root / "layers" / tmp_id
+        h = hashlib.sha256()
+        fs = src / "fs"
+        for p in sorted(fs.rglob("*")):
+            st = p.lstat()
+            h.update(str(p.relative_to(fs)).encode())
+            h.update(f"{st.st_mode}:{st.st_size}".encode())
+        lid = h.hexdigest()[:24]
+        dst = self.root / "layers" / lid
+        if dst.exists():
+            shutil.rmtree(src)
+        else:
+            os.replace(src, dst)
+        return lid
+
+    def layer_path(self, lid: str) -> Path:
+        p = self.root / "layers" / lid / "fs"
+        if not p.is_dir():
+            raise NotFoundError(f"layer missing: {lid}")
+        return p
+
+    def lowerdirs_for(self, name: str) -> list[str]:
+        """Overlay lowerdirs, TOP-most first (ckrt joins with ':')."""
+        meta = self.get(name)
+        dirs: list[str] = []
+        cur: ImageMeta | None = meta
+        seen = set()
+        while cur is not None:
+            for lid in reversed(cur.layers):   # top layer of this image first
+                dirs.append(str(self.layer_path(lid)))
+            parent = cur.parent
+            if not parent or parent in seen:
+                break
+            seen.add(parent)
+            if parent == HOSTFS:
+                dirs.append("/")
+                break
+            cur = self.get(parent)
+        if not dirs:
+            dirs.append("/")
+        return dirs
+
+    def prune_layers(self) -> int:
+        """Remove layers referenced by no image."""
+        referenced: set[str] = set()
+        for meta in self.list():
+            referenced.update(meta.layers)
+        removed = 0
+        for d in (self.root / "layers").iterdir():
+            if d.name not in referenced:
+                shutil.rmtree(d, ignore_errors=True)
+                removed += 1
+        return removed

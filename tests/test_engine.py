@@ -1,0 +1,211 @@
+"""Engine lifecycle tests with REAL sandboxes (we run as root with
+namespace support in CI; reference analog: test/e2e against real dockerd,
+but our runtime needs no daemon so these are plain unit tests)."""
+import os
+import subprocess
+import time
+from pathlib import Path
+
+import pytest
+
+from clawker_amd import consts
+from clawker_amd.engine import Engine, SandboxSpec
+from clawker_amd.engine.images import HOSTFS, ImageStore
+from clawker_amd.engine.state import StateDB
+from clawker_amd.errors import ConflictError, NotFoundError
+
+from conftest import requires_isolation
+
+
+@pytest.fixture(autouse=True)
+def _build_native():
+    root = Path(__file__).resolve().parent.parent
+    if not (root / "native/bin/ckrt").exists() or not (root / "native/bin/ckd").exists():
+        subprocess.run(["make", "native"], cwd=root, check=True, capture_output=True)
+
+
+@pytest.fixture
+def engine(isolated_env):
+    eng = Engine()
+    yield eng
+    for info in eng.list():
+        try:
+            eng.remove(info.name, force=True)
+        except Exception:
+            pass
+    eng.close()
+
+
+def _spec(name, cmd, **kw):
+    return SandboxSpec(
+        name=f"{consts.SANDBOX_NAME_PREFIX}test.{name}",
+        hostname="sbx", autostart=True, netns=True,
+        cmd=["/bin/sh", "-c", cmd], **kw)
+
+
+@requires_isolation
+def test_echo_lifecycle(engine):
+    info = engine.create(_spec("echo", "echo hello-$FOO; exit 3", env={"FOO": "bar"}))
+    assert info.state == "created"
+    engine.start(info.name)
+    code = engine.wait(info.name, timeout_s=15)
+    assert code == 3
+    assert b"hello-bar" in engine.logs(info.name)
+    info = engine.inspect(info.name)
+    assert info.state == "exited" and info.exit_code == 3
+    engine.remove(info.name)
+    with pytest.raises(NotFoundError):
+        engine.inspect(info.name)
+
+
+@requires_isolation
+def test_isolation_properties(engine):
+    info = engine.create(_spec(
+        "iso", "hostname; echo pid1=$$; ls /dev | wc -l; cat /proc/sys/kernel/hostname"))
+    engine.start(info.name)
+    assert engine.wait(info.name, timeout_s=15) == 0
+    out = engine.logs(info.name).decode()
+    assert "sbx" in out            # UTS namespace
+    # fresh PID namespace: ckd is PID 1, the agent shell gets a tiny pid
+    pid = int(out.split("pid1=")[1].split()[0])
+    assert pid < 10
+
+
+@requires_isolation
+def test_exec_into_running_sandbox(engine):
+    info = engine.create(_spec("exec", "sleep 30"))
+    engine.start(info.name)
+    code, out, err = engine.exec(info.name, ["/bin/sh", "-c", "echo from-exec; id -u"])
+    assert code == 0
+    assert b"from-exec" in out
+    # pipeline staging: stage 2 consumes stage 1 stdout
+    with engine.client(info.name) as c:
+        code, out, _ = c.exec([
+            {"argv": ["/bin/echo", "a b c"]},
+            {"argv": ["/usr/bin/tr", "a-z", "A-Z"]},
+        ])
+    assert code == 0 and b"A B C" in out
+    engine.stop(info.name)
+    assert engine.inspect(info.name).state == "exited"
+
+
+@requires_isolation
+def test_stop_forwards_sigterm(engine):
+    info = engine.create(_spec(
+        "term", "trap 'exit 42' TERM; echo up; while true; do sleep 0.1; done"))
+    engine.start(info.name)
+    # wait for the trap to be installed
+    deadline = time.monotonic() + 10
+    while b"up" not in engine.logs(info.name) and time.monotonic() < deadline:
+        time.sleep(0.02)
+    code = engine.stop(info.name)
+    assert code == 42
+
+
+@requires_isolation
+def test_writable_layer_isolated_from_host(engine, tmp_path):
+    info = engine.create(_spec("cow", "echo data > /cow-test-file && cat /cow-test-file"))
+    engine.start(info.name)
+    assert engine.wait(info.name) == 0
+    assert not Path("/cow-test-file").exists()      # COW upper, not host /
+    upper = info.statedir / "upper" / "cow-test-file"
+    assert upper.read_text().strip() == "data"
+
+
+@requires_isolation
+def test_bind_mount_workspace(engine, tmp_path):
+    ws = tmp_path / "ws"
+    ws.mkdir()
+    (ws / "input.txt").write_text("payload")
+    from clawker_amd.engine.spec import Mount
+    spec = _spec("ws", "cat /workspace/input.txt; echo done >> /workspace/out.txt",
+                 mounts=[Mount(src=str(ws), dst="/workspace")], workdir="/workspace")
+    info = engine.create(spec)
+    engine.start(info.name)
+    assert engine.wait(info.name) == 0
+    assert b"payload" in engine.logs(info.name)
+    assert (ws / "out.txt").read_text().strip() == "done"   # rw bind
+
+
+@requires_isolation
+def test_netns_denies_egress_by_construction(engine):
+    # 203.0.113.1 is TEST-NET; with an uplink-less netns any connect fails fast
+    info = engine.create(_spec(
+        "net", "ls /sys/class/net; python3 -c \"import socket;s=socket.socket();s.settimeout(2);\n"
+               "import sys\ntry: s.connect(('203.0.113.1',80)); sys.exit(1)\n"
+               "except OSError: sys.exit(0)\""))
+    engine.start(info.name)
+    assert engine.wait(info.name, timeout_s=20) == 0
+    out = engine.logs(info.name).decode()
+    assert "eth0" not in out     # no uplink interface at all
+
+
+@requires_isolation
+def test_agent_ready_gate(engine):
+    """autostart=False: CMD runs only after the CP sends agent_ready
+    (reference: clawkerd AgentReady releasing the user CMD)."""
+    spec = _spec("gate", "echo released")
+    spec.autostart = False
+    info = engine.create(spec)
+    engine.start(info.name)
+    with engine.client(info.name) as c:
+        h = c.hello()
+        assert h["cmd_running"] is False
+        assert h["initialized"] is False
+        c.agent_initialized()
+        pid = c.agent_ready()
+        assert pid > 0
+    assert engine.wait(info.name) == 0
+    assert b"released" in engine.logs(info.name)
+    # init marker persisted in the upper layer
+    marker = info.statedir / "upper" / "var/lib/clawker/initialized"
+    assert marker.exists()
+
+
+def test_duplicate_create_conflicts(engine):
+    if not os.path.exists("/proc"):
+        pytest.skip("linux only")
+    s1 = _spec("dup", "true")
+    engine.create(s1)
+    with pytest.raises(ConflictError):
+        engine.create(_spec("dup", "true"))
+
+
+def test_unmanaged_name_rejected(engine):
+    from clawker_amd.errors import EngineError
+    with pytest.raises(EngineError):
+        engine.create(SandboxSpec(name="rogue", cmd=["true"]))
+
+
+def test_volumes(engine):
+    p, fresh = engine.ensure_volume("clawker.test.vol")
+    assert fresh and p.is_dir()
+    p2, fresh2 = engine.ensure_volume("clawker.test.vol")
+    assert p2 == p and not fresh2
+    engine.remove_volume("clawker.test.vol")
+    assert engine.db.get_volume("clawker.test.vol") is None
+
+
+def test_image_store_layers(isolated_env, tmp_path):
+    store = ImageStore()
+    lid, fs = store.new_layer_dir()
+    (fs / "etc").mkdir()
+    (fs / "etc" / "marker").write_text("layer1")
+    final = store.commit_layer(lid)
+    from clawker_amd.engine.images import ImageMeta
+    store.put(ImageMeta(name="clawker-t:base", layers=[final]))
+    lowers = store.lowerdirs_for("clawker-t:base")
+    assert lowers[-1] == "/"               # hostfs at the bottom
+    assert str(store.layer_path(final)) == lowers[0]
+    # second image derived from the first
+    lid2, fs2 = store.new_layer_dir()
+    (fs2 / "top").write_text("2")
+    final2 = store.commit_layer(lid2)
+    store.put(ImageMeta(name="clawker-t:h", layers=[final2], parent="clawker-t:base"))
+    lowers2 = store.lowerdirs_for("clawker-t:h")
+    assert [Path(l).parent.name for l in lowers2[:2]] == [final2, final]
+    # prune: removing the child keeps shared layers of the parent
+    store.remove("clawker-t:h")
+    assert store.layer_path(final)         # still there
+    with pytest.raises(Exception):
+        store.layer_path(final2)
