@@ -2,11 +2,22 @@ CXX ?= g++
 CXXFLAGS ?= -O2 -std=c++17 -Wall -Wextra -g
 COMMON := native/common/minijson.hpp native/common/util.hpp
 
-.PHONY: all native test clean
+PY ?= python3
+EXT_SUFFIX := $(shell $(PY) -c "import sysconfig; print(sysconfig.get_config_var('EXT_SUFFIX'))")
+PYBIND_INC := $(shell $(PY) -c "import pybind11; print(pybind11.get_include())")
+PY_INC := $(shell $(PY) -c "import sysconfig; print(sysconfig.get_paths()['include'])")
 
-all: native
+.PHONY: all native pymod test clean
+
+all: native pymod
 
 native: native/bin/ckrt native/bin/ckd
+
+pymod: clawker_amd/_native$(EXT_SUFFIX)
+
+clawker_amd/_native$(EXT_SUFFIX): native/pymod/native.cpp
+	$(CXX) $(CXXFLAGS) -shared -fPIC -I$(PYBIND_INC) -I$(PY_INC) \
+		-o $@ native/pymod/native.cpp
 
 native/bin/ckrt: native/ckrt/ckrt.cpp $(COMMON)
 	@mkdir -p native/bin

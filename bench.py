@@ -1,0 +1,190 @@
+#!/usr/bin/env python3
+"""clawker-amd flagship benchmark: cold-start latency of GPU-pinned agent
+sandboxes (BASELINE.json metric: "cold-start p50 (ms) + max concurrent
+GPU-pinned agent loops, 1/2/4/8 MI355X").
+
+One "step" = one full cold start of a scripted agent loop: create a
+sandbox (overlay-over-hostfs rootfs, netns firewall isolation, 1 pinned
+MI355X via /dev/kfd + renderD passthrough when a GPU is present), start
+the native runtime, connect to ckd, release the agent CMD (agent_ready =
+"CMD exec'd", the cold-start endpoint per BASELINE.md), then wait for the
+scripted agent to exit and tear the sandbox down. The reported metric
+value is the p50 of the create->CMD-exec'd latency (max over ranks);
+ms_per_step additionally includes the agent run + teardown inside the
+timed region.
+
+Launch: `python bench.py --gpus N --steps K --warmup W`; for N>1 the
+driver uses torch.distributed.run with one rank per GPU — each rank pins
+its LOCAL_RANK'th MI355X and runs an independent agent loop (weak scaling).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent
+sys.path.insert(0, str(REPO))
+
+
+def setup_rank_isolation(rank: int) -> Path:
+    base = Path(os.environ.get("TMPDIR", "/tmp")) / f"clawker-bench-r{rank}"
+    for var, sub in [
+        ("CLAWKER_CONFIG_DIR", "config"), ("CLAWKER_DATA_DIR", "data"),
+        ("CLAWKER_STATE_DIR", "state"), ("CLAWKER_RUNTIME_DIR", "run"),
+        ("CLAWKER_IMAGE_DIR", "images"), ("CLAWKER_SANDBOX_DIR", "sandboxes"),
+        ("CLAWKER_VOLUME_DIR", "volumes"),
+    ]:
+        d = base / sub
+        d.mkdir(parents=True, exist_ok=True)
+        os.environ[var] = str(d)
+    return base
+
+
+TORCH_PAYLOAD = (
+    "import torch; assert torch.cuda.is_available(); "
+    "x = torch.randn(256, 256, device='cuda', requires_grad=True); "
+    "y = (x @ x).sum(); y.backward(); "
+    "torch.cuda.synchronize(); print('TORCH_OK', flush=True)"
+)
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--payload", choices=["echo", "torch"], default="echo",
+                    help="scripted agent body: echo (pure orchestration) or a "
+                         "tiny torch fwd+bwd on the pinned GPU")
+    ap.add_argument("--no-firewall", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    import torch
+    have_gpu = torch.cuda.is_available()
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        backend = "nccl" if have_gpu else "gloo"
+        if have_gpu:
+            torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        tdist.init_process_group(backend=backend)
+
+    setup_rank_isolation(rank)
+
+    # project context: synthetic workspace + default config, firewall on
+    ws = Path(os.environ["CLAWKER_STATE_DIR"]) / "workspace"
+    ws.mkdir(parents=True, exist_ok=True)
+    (ws / ".clawker.yaml").write_text(
+        "project: bench\nworkspace:\n  share_volume: false\n")
+    (ws / "task.txt").write_text("synthetic agent task\n")
+
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+
+    cfg = load_config(ws)
+    orch = Orchestrator(cfg)
+
+    use_gpu = have_gpu and args.payload in ("echo", "torch")
+    n_gpu_per_agent = 1 if have_gpu else 0
+    if args.payload == "torch" and have_gpu:
+        cmd = ["python3", "-c", TORCH_PAYLOAD]
+    else:
+        cmd = ["/bin/sh", "-c", "echo AGENT_DONE"]
+
+    firewall = not args.no_firewall
+
+    def one_cold_start(i: int) -> float:
+        name = f"clawker.bench.r{rank}s{i}"
+        t0 = time.perf_counter()
+        info = orch.run(RunOptions(
+            agent=f"r{rank}s{i}", name=name, cmd=cmd, autostart=False,
+            gpus=n_gpu_per_agent,
+            gpu_indices=[local_rank] if n_gpu_per_agent else None,
+            firewall=firewall, workspace=ws))
+        with orch.client(name) as c:
+            c.agent_ready()          # <- agent CMD exec'd: cold-start endpoint
+            t1 = time.perf_counter()
+        code = orch.engine.wait(name, timeout_s=120)
+        logs = orch.engine.logs(name)
+        orch.teardown(name, force=True)
+        if code != 0 or (b"AGENT_DONE" not in logs and b"TORCH_OK" not in logs):
+            raise RuntimeError(
+                f"agent loop failed: exit={code} logs={logs[-400:]!r}")
+        return (t1 - t0) * 1000.0
+
+    for i in range(args.warmup):
+        one_cold_start(i)
+
+    if dist:
+        dist.barrier()
+    if have_gpu:
+        torch.cuda.synchronize()
+    t_start = time.perf_counter()
+    lats = [one_cold_start(1000 + i) for i in range(args.steps)]
+    if have_gpu:
+        torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    t_end = time.perf_counter()
+
+    total_ms = (t_end - t_start) * 1000.0
+    p50 = statistics.median(lats)
+    p95 = sorted(lats)[max(0, int(len(lats) * 0.95) - 1)]
+
+    if dist:
+        vals = [None] * world
+        dist.all_gather_object(vals, {"p50": p50, "p95": p95, "total_ms": total_ms})
+        dist.barrier()
+    else:
+        vals = [{"p50": p50, "p95": p95, "total_ms": total_ms}]
+
+    if rank == 0:
+        # whole-job aggregate for a time-like metric: the slowest rank
+        agg_p50 = max(v["p50"] for v in vals)
+        agg_p95 = max(v["p95"] for v in vals)
+        ms_per_step = max(v["total_ms"] for v in vals) / args.steps
+        result = {
+            "metric": "cold-start p50 (ms) + max concurrent GPU-pinned agent loops, 1/2/4/8 MI355X",
+            "value": round(agg_p50, 3),
+            "unit": "ms",
+            "n_gpus": world if world > 1 else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "n/a",
+            "data": "synthetic",
+            "config": {
+                "model": "scripted-agent-loop (" + args.payload + ")",
+                "global_batch": world if world > 1 else args.gpus,
+                "seq_len": 0,
+                "parallelism": f"fanout{world if world > 1 else args.gpus}x1gpu",
+                "firewall": firewall,
+                "gpu_pinned": bool(n_gpu_per_agent),
+                "p95_ms": round(agg_p95, 3),
+                "concurrent_loops": world if world > 1 else 1,
+            },
+        }
+        print(json.dumps(result))
+    if dist:
+        dist.destroy_process_group()
+    orch.close()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

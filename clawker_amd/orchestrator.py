@@ -1,0 +1,198 @@
+"""The run path: config -> GPU allocation -> sandbox spec -> create/start.
+
+Reference analog: internal/cmd/container/shared/container_create.go
+(CreateContainer :1714) + container_start.go Bootstrap* — the engine-facing
+orchestration that `run`/`create`/`start` share. The MI355X-first parts:
+free-GPU allocation with 1:1 pinning, /dev/kfd + renderD passthrough,
+HBM budget env, deny-by-default netns keyed to security.firewall.
+"""
+from __future__ import annotations
+
+import shutil
+from dataclasses import dataclass, field
+from pathlib import Path
+
+from . import consts
+from .config import Config
+from .engine import CkdClient, Engine, SandboxSpec
+from .engine.images import HOSTFS
+from .engine.spec import Device, Mount
+from .errors import ClawkerError, ConflictError
+from .gpu import GPUAllocator
+from .logger import get as get_logger
+
+log = get_logger("orchestrator")
+
+
+@dataclass
+class RunOptions:
+    agent: str = "agent"
+    image: str = ""                  # "" => project image if built, else hostfs
+    cmd: list[str] = field(default_factory=list)
+    tty: bool = False
+    gpus: int | None = None          # None => project gpu.count
+    gpu_indices: list[int] | None = None
+    hbm_gb: int | None = None
+    workspace: Path | None = None
+    workspace_mode: str = ""         # "" => project setting (bind|snapshot)
+    env: dict = field(default_factory=dict)
+    user: str | None = None
+    workdir: str = ""
+    firewall: bool | None = None     # None => project security.firewall
+    autostart: bool = True           # False once the CP drives init/boot plans
+    mem_bytes: int = 0
+    pids_max: int = 4096
+    mounts: list[Mount] = field(default_factory=list)
+    labels: dict = field(default_factory=dict)
+    name: str = ""                   # override computed sandbox name
+
+
+class Orchestrator:
+    def __init__(self, cfg: Config, engine: Engine | None = None,
+                 allocator: GPUAllocator | None = None):
+        self.cfg = cfg
+        self.engine = engine or Engine()
+        self.allocator = allocator or GPUAllocator(
+            self.engine.db, reserve=cfg.settings.gpu.reserve)
+
+    def close(self) -> None:
+        self.engine.close()
+
+    # ------------------------------------------------------------------ run --
+    def create(self, opts: RunOptions):
+        """Create (not start) a sandbox from options + project config.
+        Returns SandboxInfo. GPU allocation happens here and is released
+        on create failure (reference: createScope reclaim-on-failure,
+        container_create.go:1820)."""
+        proj = self.cfg.project
+        name = opts.name or self.cfg.sandbox_name(opts.agent)
+
+        # GPU allocation (reclaim stale owners first)
+        live = {i.name for i in self.engine.list()}
+        self.allocator.reclaim_stale(live)
+        n_gpus = opts.gpus if opts.gpus is not None else proj.gpu.count
+        gpu_indices: list[int] = []
+        devices: list[Device] = []
+        if n_gpus > 0:
+            gpu_indices = self.allocator.allocate(
+                name, n_gpus, prefer_xgmi_adjacent=proj.gpu.prefer_xgmi_adjacent,
+                explicit=opts.gpu_indices)
+            devices.append(Device(path=consts.KFD_DEV))
+            for idx in gpu_indices:
+                for p in self.allocator.inventory.get(idx).device_paths():
+                    devices.append(Device(path=p))
+
+        try:
+            return self._create_inner(opts, proj, name, gpu_indices, devices)
+        except BaseException:
+            if gpu_indices:
+                self.allocator.release(name)
+            raise
+
+    def _create_inner(self, opts: RunOptions, proj, name: str,
+                      gpu_indices: list[int], devices: list[Device]):
+        firewall = opts.firewall if opts.firewall is not None else proj.security.firewall
+
+        # -- workspace mounts (reference: internal/workspace SetupMounts) ----
+        mounts: list[Mount] = []
+        ws_mode = opts.workspace_mode or proj.workspace.mode
+        ws_src = opts.workspace or (
+            self.cfg.workspace_path() if self.cfg.project_root else None)
+        ws_dst = proj.workspace.mount or "/workspace"
+        if ws_src is not None:
+            if ws_mode == "snapshot":
+                vol_name = f"{name}-snapshot"
+                vol_path, fresh = self.engine.ensure_volume(
+                    vol_name, {consts.MANAGED_LABEL: "true"})
+                if fresh:
+                    shutil.copytree(ws_src, vol_path, dirs_exist_ok=True, symlinks=True)
+                mounts.append(Mount(src=str(vol_path), dst=ws_dst))
+            else:
+                mounts.append(Mount(src=str(ws_src), dst=ws_dst))
+        if proj.workspace.share_volume and self.cfg.project_slug:
+            share, _ = self.engine.ensure_volume(
+                f"clawker.{self.cfg.project_slug}.share", {consts.MANAGED_LABEL: "true"})
+            mounts.append(Mount(src=str(share), dst="/share"))
+        mounts.extend(opts.mounts)
+
+        # -- env contract (SURVEY.md A.1 + GPU vars) -------------------------
+        env: dict[str, str] = {
+            consts.ENV_PROJECT: self.cfg.project_slug,
+            consts.ENV_AGENT: opts.agent,
+            consts.ENV_WORKSPACE_MODE: ws_mode,
+            consts.ENV_WORKSPACE_SOURCE: str(ws_src or ""),
+            consts.ENV_VERSION: "0.1.0",
+            consts.ENV_FIREWALL: "1" if firewall else "0",
+        }
+        if gpu_indices:
+            env[consts.ENV_GPU_INDEX] = ",".join(str(i) for i in gpu_indices)
+            hbm = opts.hbm_gb if opts.hbm_gb is not None else proj.gpu.hbm_gb
+            if hbm and hbm > 0:
+                env["CLAWKER_HBM_GB"] = str(hbm)
+                pct = max(1, min(100, round(hbm * 100 / 288)))
+                env["GPU_MAX_ALLOC_PERCENT"] = str(pct)
+        env.update(proj.agent.env)
+        env.update(opts.env)
+
+        image = opts.image
+        if not image:
+            candidate = self.cfg.image_name()
+            image = candidate if self.engine.images.exists(candidate) else HOSTFS
+
+        cmd = opts.cmd or proj.agent.cmd
+        user = opts.user if opts.user is not None else ""
+        workdir = opts.workdir or proj.agent.workdir or (ws_dst if ws_src else "/")
+
+        labels = {
+            consts.MANAGED_LABEL: "true",
+            consts.PROJECT_LABEL: self.cfg.project_slug,
+            consts.AGENT_LABEL: opts.agent,
+            consts.HARNESS_LABEL: proj.agent.harness,
+            consts.GPU_LABEL: ",".join(str(i) for i in gpu_indices),
+        }
+        labels.update(opts.labels)
+
+        spec = SandboxSpec(
+            name=name,
+            hostname=f"{self.cfg.project_slug or 'clawker'}-{opts.agent}"[:63],
+            netns=firewall,
+            tty=opts.tty,
+            autostart=opts.autostart,
+            mounts=mounts,
+            devices=devices,
+            mem_bytes=opts.mem_bytes,
+            pids_max=opts.pids_max,
+            env=env,
+            user=user,
+            workdir=workdir,
+            cmd=list(cmd),
+            labels=labels,
+        )
+        return self.engine.create(spec, image=image)
+
+    def start(self, name: str):
+        return self.engine.start(name)
+
+    def run(self, opts: RunOptions):
+        """create + start; returns SandboxInfo (agent CMD is already
+        spawning if autostart, else waiting for agent_ready)."""
+        info = self.create(opts)
+        try:
+            return self.engine.start(info.name)
+        except BaseException:
+            self.teardown(info.name, force=True)
+            raise
+
+    def teardown(self, name: str, force: bool = False) -> None:
+        try:
+            self.engine.remove(name, force=force)
+        finally:
+            self.allocator.release(name)
+            # snapshot volume cleanup
+            try:
+                self.engine.remove_volume(f"{name}-snapshot")
+            except ClawkerError:
+                pass
+
+    def client(self, name: str, **kw) -> CkdClient:
+        return self.engine.client(name, **kw)

@@ -1,0 +1,129 @@
+"""MI355X GPU tests (run via gpurun on a real box). Everything here is
+marked gpu; the driver runs `pytest -m gpu` on hardware at round end."""
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _warm_torch():
+    # first `import torch` on a fresh box pages in the image; do it once
+    # host-side so sandbox imports hit the page cache
+    import torch  # noqa: F401
+
+
+@pytest.fixture
+def orch(isolated_env):
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator
+    ws = isolated_env / "ws"
+    ws.mkdir(exist_ok=True)
+    (ws / ".clawker.yaml").write_text("project: gputest\n")
+    o = Orchestrator(load_config(ws))
+    yield o
+    for info in o.engine.list():
+        try:
+            o.teardown(info.name, force=True)
+        except Exception:
+            pass
+    o.close()
+
+
+def test_inventory_detects_amdgpu():
+    from clawker_amd.gpu import GPUInventory
+    inv = GPUInventory.detect()
+    assert len(inv) >= 1
+    d = inv.get(0)
+    assert d.render_path.startswith("/dev/dri/renderD")
+    assert Path(d.render_path).exists()
+    assert Path("/dev/kfd").exists()
+    # MI355X: 288 GB HBM3E
+    assert d.vram_total > 200 * 2**30, f"vram {d.vram_total}"
+
+
+def test_native_sampler_reads_live_telemetry():
+    from clawker_amd.monitor import RocmSampler
+    s = RocmSampler()
+    samples = s.sample()
+    assert samples, "no GPUs sampled"
+    row = samples[0]
+    assert row.vram_total > 0
+    assert 0 <= row.busy_pct <= 100
+    # the native extension must actually be loaded (no eager fallback)
+    assert "clawker_amd._native" in sys.modules
+
+
+def test_gpu_pinned_sandbox_runs_torch(orch):
+    from clawker_amd.orchestrator import RunOptions
+    payload = (
+        "import torch; assert torch.cuda.is_available(); "
+        "n = torch.cuda.device_count(); assert n == 1, f'visible={n}'; "
+        "x = torch.randn(256, 256, device='cuda', requires_grad=True); "
+        "(x @ x).sum().backward(); torch.cuda.synchronize(); "
+        "print('GPU_SANDBOX_OK', flush=True)")
+    name = "clawker.gputest.t1"
+    orch.run(RunOptions(agent="t1", name=name, gpus=1, autostart=True,
+                        cmd=["python3", "-c", payload], firewall=True))
+    code = orch.engine.wait(name, timeout_s=300)
+    logs = orch.engine.logs(name)
+    assert code == 0, logs[-800:]
+    assert b"GPU_SANDBOX_OK" in logs
+
+
+def test_device_isolation_only_allocated_render_node(orch):
+    from clawker_amd.gpu import GPUInventory
+    from clawker_amd.orchestrator import RunOptions
+    inv = GPUInventory.detect()
+    name = "clawker.gputest.iso"
+    orch.run(RunOptions(
+        agent="iso", name=name, gpus=1, autostart=True, firewall=True,
+        cmd=["/bin/sh", "-c", "ls /dev/dri/ | grep -c renderD; ls /dev/kfd"]))
+    code = orch.engine.wait(name, timeout_s=60)
+    logs = orch.engine.logs(name).decode()
+    assert code == 0, logs
+    visible = int(logs.splitlines()[0])
+    assert visible == 1, f"sandbox sees {visible} render nodes (host has {len(inv)})"
+    assert "/dev/kfd" in logs
+
+
+def test_gpu_allocation_ledger_and_release(orch):
+    from clawker_amd.gpu import GPUInventory
+    from clawker_amd.orchestrator import RunOptions
+    inv = GPUInventory.detect()
+    name = "clawker.gputest.alloc"
+    orch.run(RunOptions(agent="alloc", name=name, gpus=1, autostart=True,
+                        cmd=["/bin/sh", "-c", "sleep 5"], firewall=True))
+    allocs = orch.allocator.allocations()
+    assert list(allocs.values()) == [name]
+    if len(inv) == 1:
+        # second 1-GPU request must fail while the only GPU is held
+        from clawker_amd.gpu import GPUAllocationError
+        with pytest.raises(GPUAllocationError):
+            orch.create(RunOptions(agent="second", name="clawker.gputest.second",
+                                   gpus=1, cmd=["true"]))
+    orch.teardown(name, force=True)
+    assert orch.allocator.allocations() == {}
+
+
+def test_bench_single_gpu_quick():
+    """bench.py contract: one JSON line, sane fields, on-GPU cold starts."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"), "--steps", "3", "--warmup", "1"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(REPO))
+    assert r.returncode == 0, r.stderr[-2000:]
+    import json
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["higher_is_better"] is False
+    assert out["value"] > 0 and out["ms_per_step"] > 0
+    assert out["config"]["gpu_pinned"] is True

@@ -1,0 +1,123 @@
+"""Orchestrator tests on CPU (no GPU allocation; real sandboxes)."""
+import json
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+from conftest import requires_isolation
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+@pytest.fixture
+def proj(isolated_env, tmp_path):
+    ws = tmp_path / "proj"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text(
+        "project: otest\nagent:\n  env:\n    FROM_PROJECT: yes1\n")
+    (ws / "data.txt").write_text("payload")
+    return ws
+
+
+@pytest.fixture
+def orch(proj):
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator
+    o = Orchestrator(load_config(proj))
+    yield o
+    for info in o.engine.list():
+        try:
+            o.teardown(info.name, force=True)
+        except Exception:
+            pass
+    o.close()
+
+
+@requires_isolation
+def test_run_env_contract_and_workspace(orch, proj):
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.otest.a1"
+    orch.run(RunOptions(
+        agent="a1", name=name, autostart=True,
+        cmd=["/bin/sh", "-c",
+             "echo P=$CLAWKER_PROJECT A=$CLAWKER_AGENT F=$CLAWKER_FIREWALL "
+             "E=$FROM_PROJECT; cat /workspace/data.txt; pwd"]))
+    assert orch.engine.wait(name, timeout_s=30) == 0
+    out = orch.engine.logs(name).decode()
+    assert "P=otest A=a1 F=1 E=yes1" in out
+    assert "payload" in out
+    assert "/workspace" in out        # default workdir = workspace mount
+
+
+@requires_isolation
+def test_snapshot_workspace_is_disposable(orch, proj):
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.otest.snap"
+    orch.run(RunOptions(
+        agent="snap", name=name, autostart=True, workspace_mode="snapshot",
+        cmd=["/bin/sh", "-c", "echo scribble >> /workspace/data.txt; cat /workspace/data.txt"]))
+    assert orch.engine.wait(name, timeout_s=30) == 0
+    assert b"scribble" in orch.engine.logs(name)
+    # host workspace untouched (reference: SnapshotStrategy disposable copy)
+    assert (proj / "data.txt").read_text() == "payload"
+    orch.teardown(name, force=True)
+
+
+@requires_isolation
+def test_no_firewall_shares_host_netns(orch):
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.otest.hostnet"
+    orch.run(RunOptions(agent="hostnet", name=name, autostart=True, firewall=False,
+                        cmd=["/bin/sh", "-c", "ls /sys/class/net"]))
+    assert orch.engine.wait(name, timeout_s=30) == 0
+    out = orch.engine.logs(name).decode()
+    assert "eth0" in out or "ens" in out or "enp" in out
+
+
+def test_gpu_request_fails_without_gpus(orch):
+    from clawker_amd.gpu import GPUAllocationError
+    from clawker_amd.orchestrator import RunOptions
+    if orch.allocator.inventory.devices:
+        pytest.skip("host has GPUs")
+    with pytest.raises(GPUAllocationError):
+        orch.create(RunOptions(agent="g", name="clawker.otest.g", gpus=1, cmd=["true"]))
+
+
+@requires_isolation
+def test_bench_cpu_single():
+    env = dict(os.environ)
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
+        env.pop(k, None)
+    r = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"), "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=300, env=env, cwd=str(REPO))
+    assert r.returncode == 0, r.stderr[-3000:]
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["unit"] == "ms"
+    assert out["steps"] == 2
+    assert out["data"] == "synthetic"
+    assert out["config"]["gpu_pinned"] is False      # CPU box
+
+
+@requires_isolation
+def test_bench_distributed_gloo_world2():
+    """The driver launches bench via torch.distributed.run for N>1;
+    exercise that path with gloo on CPU (world_size=2)."""
+    env = dict(os.environ)
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
+        env.pop(k, None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29517", str(REPO / "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "0"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=str(REPO))
+    assert r.returncode == 0, (r.stdout[-2000:], r.stderr[-3000:])
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["n_gpus"] == 2
+    assert out["config"]["concurrent_loops"] == 2

@@ -1,0 +1,99 @@
+#!/usr/bin/env python3
+"""Collect a rocprofv3 kernel-trace profile of a torch workload running
+INSIDE a clawker-amd sandbox (proves the monitoring/profiling north star:
+rocprof works through our /dev/kfd + renderD passthrough), with a host-side
+fallback profile of the identical payload.
+
+Usage: python tools/profile_sandbox.py <output_dir>
+"""
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+sys.path.insert(0, str(REPO))
+
+PAYLOAD = (
+    "import torch; assert torch.cuda.is_available(); "
+    "a = torch.randn(4096, 4096, device='cuda', dtype=torch.bfloat16); "
+    "b = torch.randn(4096, 4096, device='cuda', dtype=torch.bfloat16); "
+    "import time\n"
+    "for _ in range(3): c = a @ b\n"
+    "torch.cuda.synchronize(); t0 = time.time()\n"
+    "for _ in range(20): c = a @ b\n"
+    "torch.cuda.synchronize(); dt = time.time() - t0\n"
+    "fl = 20 * 2 * 4096**3 / dt / 1e12\n"
+    "print(f'MATMUL_TFLOPS {fl:.1f}', flush=True)"
+)
+
+
+def main() -> int:
+    out_dir = Path(sys.argv[1] if len(sys.argv) > 1 else "gpurun_out/prof").resolve()
+    out_dir.mkdir(parents=True, exist_ok=True)
+
+    base = Path("/tmp/clawker-prof")
+    for var, sub in [
+        ("CLAWKER_CONFIG_DIR", "config"), ("CLAWKER_DATA_DIR", "data"),
+        ("CLAWKER_STATE_DIR", "state"), ("CLAWKER_RUNTIME_DIR", "run"),
+        ("CLAWKER_IMAGE_DIR", "images"), ("CLAWKER_SANDBOX_DIR", "sandboxes"),
+        ("CLAWKER_VOLUME_DIR", "volumes"),
+    ]:
+        d = base / sub
+        d.mkdir(parents=True, exist_ok=True)
+        os.environ[var] = str(d)
+
+    from clawker_amd.config import load_config
+    from clawker_amd.engine.spec import Mount
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+
+    ws = base / "ws"
+    ws.mkdir(exist_ok=True)
+    (ws / ".clawker.yaml").write_text("project: prof\n")
+    orch = Orchestrator(load_config(ws))
+    name = "clawker.prof.agent"
+
+    # in-sandbox rocprof: output bound to the host out_dir
+    cmd = ["/bin/sh", "-c",
+           "cd /tmp && TMPDIR=/tmp rocprofv3 --kernel-trace --stats "
+           "-d /profout/sandbox -- python3 -c \"" + PAYLOAD.replace('"', '\\"') + "\""]
+    ok = False
+    try:
+        try:
+            orch.teardown(name, force=True)
+        except Exception:
+            pass
+        orch.run(RunOptions(
+            agent="prof", name=name, gpus=1, gpu_indices=[0], autostart=True,
+            cmd=cmd, firewall=True,
+            mounts=[Mount(src=str(out_dir), dst="/profout")]))
+        code = orch.engine.wait(name, timeout_s=240)
+        logs = orch.engine.logs(name).decode(errors="replace")
+        (out_dir / "sandbox_console.log").write_text(logs)
+        print(f"in-sandbox rocprof exit={code}")
+        print(logs[-2000:])
+        ok = code == 0 and "MATMUL_TFLOPS" in logs
+    finally:
+        try:
+            orch.teardown(name, force=True)
+        except Exception:
+            pass
+        orch.close()
+
+    if not ok:
+        print("in-sandbox rocprof failed; collecting host-side fallback")
+        r = subprocess.run(
+            ["rocprofv3", "--kernel-trace", "--stats", "-d", str(out_dir / "host"),
+             "--", sys.executable, "-c", PAYLOAD],
+            cwd="/tmp", env=dict(os.environ, TMPDIR="/tmp"),
+            capture_output=True, text=True, timeout=240)
+        (out_dir / "host_rocprof.log").write_text(r.stdout + "\n" + r.stderr)
+        print(r.stdout[-2000:])
+        return 1 if r.returncode else 0
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
