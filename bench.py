@@ -173,6 +173,7 @@ def main() -> int:
                 "global_batch": world if world > 1 else args.gpus,
                 "seq_len": 0,
                 "parallelism": f"fanout{world if world > 1 else args.gpus}x1gpu",
+                "isolation": orch.engine.backend,
                 "firewall": firewall,
                 "gpu_pinned": bool(n_gpu_per_agent),
                 "p95_ms": round(agg_p95, 3),

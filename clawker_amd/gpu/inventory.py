@@ -71,6 +71,11 @@ class GPUInventory:
             if _read(dev / "vendor") != AMD_VENDOR:
                 continue
             minor = int(r.name[7:])
+            # containerized hosts expose the full host sysfs but only a
+            # subset of /dev/dri nodes — a GPU without an accessible render
+            # node is unusable, so it is not inventory
+            if not Path(f"/dev/dri/renderD{minor}").exists():
+                continue
             # find the cardN sharing this PCI device
             card = -1
             pci = ""

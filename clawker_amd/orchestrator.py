@@ -91,7 +91,12 @@ class Orchestrator:
 
     def _create_inner(self, opts: RunOptions, proj, name: str,
                       gpu_indices: list[int], devices: list[Device]):
+        backend = self.engine.backend
         firewall = opts.firewall if opts.firewall is not None else proj.security.firewall
+        # netns-based egress denial needs the ns backend; degrade honestly
+        effective_firewall = firewall and backend == "ns"
+        if firewall and not effective_firewall:
+            log.warn("firewall_unavailable", sandbox=name, backend=backend)
 
         # -- workspace mounts (reference: internal/workspace SetupMounts) ----
         mounts: list[Mount] = []
@@ -99,6 +104,7 @@ class Orchestrator:
         ws_src = opts.workspace or (
             self.cfg.workspace_path() if self.cfg.project_root else None)
         ws_dst = proj.workspace.mount or "/workspace"
+        ws_effective = ws_src        # agent-visible workspace path
         if ws_src is not None:
             if ws_mode == "snapshot":
                 vol_name = f"{name}-snapshot"
@@ -107,8 +113,10 @@ class Orchestrator:
                 if fresh:
                     shutil.copytree(ws_src, vol_path, dirs_exist_ok=True, symlinks=True)
                 mounts.append(Mount(src=str(vol_path), dst=ws_dst))
+                ws_effective = vol_path if backend == "proc" else Path(ws_dst)
             else:
                 mounts.append(Mount(src=str(ws_src), dst=ws_dst))
+                ws_effective = ws_src if backend == "proc" else Path(ws_dst)
         if proj.workspace.share_volume and self.cfg.project_slug:
             share, _ = self.engine.ensure_volume(
                 f"clawker.{self.cfg.project_slug}.share", {consts.MANAGED_LABEL: "true"})
@@ -122,10 +130,14 @@ class Orchestrator:
             consts.ENV_WORKSPACE_MODE: ws_mode,
             consts.ENV_WORKSPACE_SOURCE: str(ws_src or ""),
             consts.ENV_VERSION: "0.1.0",
-            consts.ENV_FIREWALL: "1" if firewall else "0",
+            consts.ENV_FIREWALL: "1" if effective_firewall else "0",
         }
         if gpu_indices:
             env[consts.ENV_GPU_INDEX] = ",".join(str(i) for i in gpu_indices)
+            if backend == "proc":
+                # no device-node isolation without a mount namespace: select
+                # the pinned GPUs via the HSA runtime's visibility filter
+                env[consts.ENV_ROCR_VISIBLE] = ",".join(str(i) for i in gpu_indices)
             hbm = opts.hbm_gb if opts.hbm_gb is not None else proj.gpu.hbm_gb
             if hbm and hbm > 0:
                 env["CLAWKER_HBM_GB"] = str(hbm)
@@ -141,7 +153,8 @@ class Orchestrator:
 
         cmd = opts.cmd or proj.agent.cmd
         user = opts.user if opts.user is not None else ""
-        workdir = opts.workdir or proj.agent.workdir or (ws_dst if ws_src else "/")
+        default_workdir = str(ws_effective) if ws_src is not None else "/"
+        workdir = opts.workdir or proj.agent.workdir or default_workdir
 
         labels = {
             consts.MANAGED_LABEL: "true",
@@ -155,7 +168,7 @@ class Orchestrator:
         spec = SandboxSpec(
             name=name,
             hostname=f"{self.cfg.project_slug or 'clawker'}-{opts.agent}"[:63],
-            netns=firewall,
+            netns=effective_firewall,
             tty=opts.tty,
             autostart=opts.autostart,
             mounts=mounts,

@@ -54,12 +54,16 @@ using ck::warn;
 
 namespace {
 
-constexpr const char* kCtlSock = "/run/clawker/ctl.sock";
-constexpr const char* kConsoleLog = "/run/clawker/console.log";
-constexpr const char* kReadyDir = "/var/run/clawker";
-constexpr const char* kReadyFile = "/var/run/clawker/ready";
-constexpr const char* kInitMarkerDir = "/var/lib/clawker";
-constexpr const char* kInitMarker = "/var/lib/clawker/initialized";
+// Paths come from the spec's "paths" object so ckd works under both
+// isolation backends: "ns" (rundir bind-mounted at /run/clawker inside the
+// sandbox) and "proc" (no mount namespace available — e.g. a restricted
+// container host — so rundir is the per-sandbox host directory itself).
+std::string g_rundir = "/run/clawker";
+std::string g_marker = "/var/lib/clawker/initialized";
+
+std::string ctl_sock() { return g_rundir + "/ctl.sock"; }
+std::string console_log() { return g_rundir + "/console.log"; }
+std::string ready_file() { return g_rundir + "/ready"; }
 
 mj::Value g_spec;
 int g_selfpipe[2];    // SIGCHLD -> poll wakeup
@@ -429,7 +433,7 @@ void handle_frame(Client& cl, const mj::Value& req) {
   if (t == "hello") {
     mj::Value r;
     r.set("t", "hello")
-     .set("initialized", ck::exists(kInitMarker))
+     .set("initialized", ck::exists(g_marker))
      .set("cmd_running", g_agent.running)
      .set("pid", (int64_t)(g_agent.running ? g_agent.pid : -1))
      .set("version", "0.1.0");
@@ -441,8 +445,9 @@ void handle_frame(Client& cl, const mj::Value& req) {
     r.set("t", "ready_ack").set("pid", (int64_t)g_agent.pid);
     ck::send_frame(cl.fd, r);
   } else if (t == "agent_initialized") {
-    ck::mkdirs(kInitMarkerDir);
-    ck::write_file(kInitMarker, std::to_string(time(nullptr)));
+    size_t slash = g_marker.rfind('/');
+    if (slash != std::string::npos) ck::mkdirs(g_marker.substr(0, slash));
+    ck::write_file(g_marker, std::to_string(time(nullptr)));
     mj::Value r; r.set("t", "ok");
     ck::send_frame(cl.fd, r);
   } else if (t == "exec") {
@@ -471,7 +476,7 @@ void handle_frame(Client& cl, const mj::Value& req) {
     mj::Value r;
     r.set("t", "status").set("cmd_running", g_agent.running)
      .set("exit_code", (int64_t)g_agent.exit_code)
-     .set("initialized", ck::exists(kInitMarker));
+     .set("initialized", ck::exists(g_marker));
     ck::send_frame(cl.fd, r);
   } else {
     mj::Value r; r.set("t", "error").set("msg", "unknown command: " + t);
@@ -514,6 +519,9 @@ int main() {
   const char* spec_path = getenv("CKD_SPEC");
   if (!spec_path) die("CKD_SPEC not set");
   g_spec = mj::parse(ck::read_file(spec_path));
+  const mj::Value& paths = g_spec["paths"];
+  if (paths.has("rundir")) g_rundir = paths["rundir"].as_str();
+  if (paths.has("marker")) g_marker = paths["marker"].as_str();
 
   if (pipe2(g_selfpipe, O_CLOEXEC | O_NONBLOCK) != 0) die("selfpipe");
   struct sigaction sa{};
@@ -534,15 +542,15 @@ int main() {
   for (int sig : {SIGTERM, SIGINT, SIGHUP, SIGQUIT, SIGUSR1, SIGUSR2})
     sigaction(sig, &fsa, nullptr);
 
-  int listen_fd = ck::unix_listen(kCtlSock);
-  if (listen_fd < 0) die("listen %s", kCtlSock);
+  int listen_fd = ck::unix_listen(ctl_sock());
+  if (listen_fd < 0) die("listen %s", ctl_sock().c_str());
   fcntl(listen_fd, F_SETFL, O_NONBLOCK);
 
-  g_console_log = open(kConsoleLog, O_WRONLY | O_CREAT | O_APPEND | O_CLOEXEC, 0600);
+  g_console_log = open(console_log().c_str(),
+                       O_WRONLY | O_CREAT | O_APPEND | O_CLOEXEC, 0600);
 
   // ready file: the HEALTHCHECK analog (reference: Dockerfile.base.tmpl:245)
-  ck::mkdirs(kReadyDir);
-  ck::write_file(kReadyFile, "1");
+  ck::write_file(ready_file(), "1");
 
   if (g_spec["autostart"].as_bool(false)) spawn_agent(mj::Value());
 
