@@ -308,7 +308,7 @@ class Engine:
                 return -1
             if deadline is not None and time.monotonic() > deadline:
                 raise EngineError("wait", f"timeout waiting for {name}")
-            time.sleep(0.02)
+            time.sleep(0.005)
 
     # ------------------------------------------------------------ remove ----
     def remove(self, name: str, force: bool = False) -> None:

@@ -577,7 +577,7 @@ int main() {
       if (kv.second.err >= 0) { exec_fds.push_back({kv.first, 2}); pfds.push_back({kv.second.err, POLLIN, 0}); }
     }
 
-    int timeout = exiting ? 50 : 1000;
+    int timeout = exiting ? 10 : 1000;
     int rc = poll(pfds.data(), pfds.size(), timeout);
     if (rc < 0 && errno != EINTR) die("poll");
 
@@ -647,7 +647,14 @@ int main() {
       exiting = true;
       exit_deadline_ms = now_ms() + 500;
     }
-    if (exiting && now_ms() >= exit_deadline_ms)
-      return g_agent.exit_code < 0 ? 0 : g_agent.exit_code;
+    if (exiting) {
+      // leave as soon as the orphan drain is dry (no remaining children)
+      // or after the bounded grace; pidns teardown reclaims stragglers
+      int st;
+      pid_t r = waitpid(-1, &st, WNOHANG);
+      bool no_children = (r < 0 && errno == ECHILD);
+      if (no_children || now_ms() >= exit_deadline_ms)
+        return g_agent.exit_code < 0 ? 0 : g_agent.exit_code;
+    }
   }
 }

@@ -55,10 +55,14 @@ def main() -> int:
     orch = Orchestrator(load_config(ws))
     name = "clawker.prof.agent"
 
-    # in-sandbox rocprof: output bound to the host out_dir
+    # in-sandbox rocprof: ns backend sees out_dir bound at /profout; the
+    # proc backend shares the host fs so the host path works directly
+    prof_dst = ("/profout/sandbox" if orch.engine.backend == "ns"
+                else str(out_dir / "sandbox"))
     cmd = ["/bin/sh", "-c",
            "cd /tmp && TMPDIR=/tmp rocprofv3 --kernel-trace --stats "
-           "-d /profout/sandbox -- python3 -c \"" + PAYLOAD.replace('"', '\\"') + "\""]
+           f"-d {prof_dst} "
+           "-- python3 -c \"" + PAYLOAD.replace('"', '\\"') + "\""]
     ok = False
     try:
         try:
