@@ -1,0 +1,60 @@
+"""Control-plane verbs (reference: internal/cmd/controlplane up/down/
+status/agents)."""
+from __future__ import annotations
+
+import json
+
+import click
+
+from .root import Ctx, cli, pass_factory
+
+
+@cli.group("cp")
+def cp_group():
+    """Control-plane daemon (cpd)."""
+
+
+@cp_group.command("up")
+@pass_factory
+def cp_up(ctx: Ctx):
+    cp = ctx.factory.controlplane()
+    cp.ensure_running()
+    ctx.factory.io.success("control plane ready")
+
+
+@cp_group.command("down")
+@pass_factory
+def cp_down(ctx: Ctx):
+    cp = ctx.factory.controlplane()
+    cp.auto_start = False
+    if cp.stop():
+        ctx.factory.io.success("control plane stopped")
+    else:
+        ctx.factory.io.eprint("control plane not running")
+
+
+@cp_group.command("status")
+@pass_factory
+def cp_status(ctx: Ctx):
+    cp = ctx.factory.controlplane()
+    cp.auto_start = False
+    if not cp.running():
+        ctx.factory.io.print(json.dumps({"running": False}))
+        return
+    ctx.factory.io.print(json.dumps({"running": True, **cp.status()}, indent=1))
+
+
+@cp_group.command("agents")
+@pass_factory
+def cp_agents(ctx: Ctx):
+    cp = ctx.factory.controlplane()
+    ctx.factory.io.print(json.dumps(cp.agents(), indent=1))
+
+
+@cp_group.command("events")
+@click.option("-n", type=int, default=50, show_default=True)
+@pass_factory
+def cp_events(ctx: Ctx, n):
+    cp = ctx.factory.controlplane()
+    for ev in cp.events(n):
+        ctx.factory.io.print(json.dumps(ev))

@@ -1,0 +1,127 @@
+"""Firewall verbs (reference: internal/cmd/firewall — status/list/add/
+remove/reload/enable/disable/bypass backed by the 13 AdminService RPCs;
+here backed by the rules store + control-plane daemon)."""
+from __future__ import annotations
+
+import json
+import time
+
+import click
+
+from ..config.schema import EgressRule
+from ..errors import ClawkerError
+from ..firewall import EgressRulesStore, IdentityAllocator
+from .root import Ctx, cli, pass_factory
+
+
+@cli.group("firewall")
+def firewall_group():
+    """Deny-by-default egress policy."""
+
+
+def _store() -> EgressRulesStore:
+    return EgressRulesStore()
+
+
+@firewall_group.command("list")
+@click.option("--format", "fmt", default="")
+@pass_factory
+def fw_list(ctx: Ctx, fmt):
+    f = ctx.factory
+    rules = _store().list()
+    idents = IdentityAllocator()
+    if fmt == "json":
+        from ..storage.store import to_plain
+        f.io.print(json.dumps([to_plain(r) for r in rules], indent=1))
+        return
+    from rich.table import Table
+    t = Table(box=None, pad_edge=False)
+    for c in ("DST", "PROTO", "PORT", "IDENTITY", "PATHS", "DENY-PATHS"):
+        t.add_column(c)
+    for r in rules:
+        ident = idents.get(r.dst)
+        t.add_row(r.dst, r.proto, str(r.port), str(ident or "-"),
+                  ",".join(r.paths) or "-", ",".join(r.deny_paths) or "-")
+    f.io.print(t)
+
+
+@firewall_group.command("add")
+@click.argument("dst")
+@click.option("--proto", type=click.Choice(["tls", "http", "tcp", "udp"]),
+              default="tls", show_default=True)
+@click.option("--port", type=int, default=443, show_default=True)
+@click.option("--path", "paths", multiple=True, help="allow-only path prefix ('~' = regex)")
+@click.option("--deny-path", "deny_paths", multiple=True)
+@pass_factory
+def fw_add(ctx: Ctx, dst, proto, port, paths, deny_paths):
+    """Allow egress to DST (domain, *.wildcard, or IP)."""
+    f = ctx.factory
+    rule = EgressRule(dst=dst, proto=proto, port=port,
+                      paths=list(paths), deny_paths=list(deny_paths))
+    changed = _store().add([rule])
+    IdentityAllocator().allocate(dst)
+    _reload_running(f)
+    f.io.success(f"{'added' if changed else 'already present'}: {rule.key()}")
+
+
+@firewall_group.command("remove")
+@click.argument("dst")
+@pass_factory
+def fw_remove(ctx: Ctx, dst):
+    f = ctx.factory
+    if not _store().remove(dst):
+        raise ClawkerError(f"no rule for: {dst}")
+    _reload_running(f)
+    f.io.success(f"removed: {dst}")
+
+
+@firewall_group.command("status")
+@pass_factory
+def fw_status(ctx: Ctx):
+    f = ctx.factory
+    rules = _store().list()
+    infos = f.engine().list()
+    enforced = [i.name for i in infos if i.state == "running"
+                and i.labels.get("dev.clawker.fw") != "off"]
+    backend = f.engine().backend
+    f.io.print(json.dumps({
+        "backend": backend,
+        "enforcement": "netns+gateway" if backend == "ns" else
+                       "unavailable (proc backend: no netns on this host)",
+        "rules": len(rules),
+        "running_sandboxes": len([i for i in infos if i.state == "running"]),
+    }, indent=1))
+
+
+@firewall_group.command("reload")
+@pass_factory
+def fw_reload(ctx: Ctx):
+    """Push current rules to every running sandbox's gateway."""
+    f = ctx.factory
+    n = _reload_running(f)
+    f.io.success(f"reloaded policy on {n} gateway(s)")
+
+
+@firewall_group.command("bypass")
+@click.option("--minutes", type=int, default=15, show_default=True)
+@pass_factory
+def fw_bypass(ctx: Ctx, minutes):
+    """Temporarily allow all egress (dead-man capped at settings
+    firewall.bypass_max_s; reference: FirewallBypass 1h cap)."""
+    f = ctx.factory
+    cap = f.config().settings.firewall.bypass_max_s
+    secs = min(minutes * 60, cap)
+    from ..controlplane.client import CPClient
+    cp = f.controlplane()
+    cp.bypass(secs)
+    f.io.warn(f"firewall BYPASSED for {secs}s (auto-restore)")
+
+
+def _reload_running(f) -> int:
+    """Signal running sandbox gateways to re-read policy (the gateway polls
+    the policy file; CP push lands in controlplane/daemon)."""
+    try:
+        cp = f.controlplane()
+        return cp.reload_policy()
+    except Exception:
+        return 0

@@ -1,0 +1,85 @@
+"""Image verbs (reference: internal/cmd/image — list/build/inspect/remove/
+prune) plus the top-level `clawker build` alias."""
+from __future__ import annotations
+
+import json
+import time
+
+import click
+
+from ..cmdutil import format_age
+from .root import Ctx, cli, pass_factory
+
+
+@cli.group("image")
+def image_group():
+    """Manage sandbox images (overlay layer stacks over hostfs)."""
+
+
+@cli.command("build")
+@click.option("--harness", default="", help="harness to bake (default: project agent.harness)")
+@click.option("--no-cache", is_flag=True, help="rebuild the base even if fresh")
+@click.option("-q", "--quiet", is_flag=True)
+@pass_factory
+def build_cmd(ctx: Ctx, harness, no_cache, quiet):
+    """Build the project image (base + harness stages)."""
+    f = ctx.factory
+    cfg = f.config(require_project=True)
+    from ..bundler import Builder
+    builder = Builder(cfg, f.engine())
+    progress = None if quiet else (lambda line: f.io.eprint(f"[dim]»[/dim] {line}"))
+    name = builder.build(harness_name=harness, no_cache=no_cache,
+                         on_progress=progress)
+    f.io.print(name)
+
+
+image_group.add_command(build_cmd, "build")
+
+
+@image_group.command("ls")
+@click.option("--format", "fmt", default="")
+@pass_factory
+def image_ls(ctx: Ctx, fmt):
+    """List images."""
+    f = ctx.factory
+    metas = f.engine().images.list()
+    if fmt == "json":
+        f.io.print(json.dumps([m.to_dict() for m in metas], indent=1))
+        return
+    from rich.table import Table
+    t = Table(box=None, pad_edge=False)
+    for col in ("NAME", "PARENT", "LAYERS", "AGE"):
+        t.add_column(col)
+    now = time.time()
+    for m in metas:
+        t.add_row(m.name, m.parent, str(len(m.layers)),
+                  format_age(now - m.created) if m.created else "-")
+    f.io.print(t)
+
+
+@image_group.command("inspect")
+@click.argument("names", nargs=-1, required=True)
+@pass_factory
+def image_inspect(ctx: Ctx, names):
+    f = ctx.factory
+    out = [f.engine().images.get(n).to_dict() for n in names]
+    f.io.print(json.dumps(out, indent=1))
+
+
+@image_group.command("rm")
+@click.argument("names", nargs=-1, required=True)
+@pass_factory
+def image_rm(ctx: Ctx, names):
+    f = ctx.factory
+    for n in names:
+        f.engine().images.remove(n)
+        f.io.print(n)
+
+
+@image_group.command("prune")
+@pass_factory
+def image_prune(ctx: Ctx):
+    """Remove layers referenced by no image."""
+    f = ctx.factory
+    n = f.engine().images.prune_layers()
+    f.io.eprint(f"removed {n} unreferenced layer(s)")

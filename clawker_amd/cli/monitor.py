@@ -1,0 +1,136 @@
+"""Monitoring verbs: live GPU + sandbox telemetry.
+
+Reference: internal/cmd/monitor (compose observability stack) and
+internal/cmd/container/stats (streamStats repaint loop). MI355X-first:
+`clawker stats` streams a live table of per-sandbox cgroup stats and
+per-GPU rocm telemetry from the zero-spawn native sampler; `clawker gpus`
+shows the inventory/allocation map; `clawker monitor up` starts the
+Prometheus exporter daemon."""
+from __future__ import annotations
+
+import json
+import time
+
+import click
+
+from .root import Ctx, cli, pass_factory
+
+
+@cli.command("gpus")
+@click.option("--format", "fmt", default="")
+@pass_factory
+def gpus_cmd(ctx: Ctx, fmt):
+    """GPU inventory and allocations."""
+    f = ctx.factory
+    orch = f.orchestrator()
+    inv = orch.allocator.inventory
+    allocs = orch.allocator.allocations()
+    rows = []
+    samples = {}
+    try:
+        from ..monitor import RocmSampler
+        for s in RocmSampler(inv).sample():
+            samples[s.index] = s
+    except Exception:
+        pass
+    for d in inv.devices:
+        s = samples.get(d.index)
+        rows.append({
+            "index": d.index, "render": d.render_path, "pci": d.pci_bus,
+            "vram_gb": round(d.vram_total / 2**30, 1),
+            "owner": allocs.get(d.index, ""),
+            "busy_pct": s.busy_pct if s else None,
+            "vram_used_gb": round(s.vram_used / 2**30, 1) if s else None,
+            "power_w": s.power_w if s else None,
+            "temp_c": s.temp_junction_c or s.temp_edge_c if s else None,
+            "xgmi_peers": d.xgmi_peers,
+        })
+    if fmt == "json":
+        f.io.print(json.dumps(rows, indent=1))
+        return
+    from rich.table import Table
+    t = Table(box=None, pad_edge=False)
+    for c in ("GPU", "RENDER", "VRAM", "BUSY", "USED", "POWER", "TEMP", "OWNER"):
+        t.add_column(c)
+    for r in rows:
+        t.add_row(str(r["index"]), r["render"], f"{r['vram_gb']}G",
+                  f"{r['busy_pct']:.0f}%" if r["busy_pct"] is not None else "-",
+                  f"{r['vram_used_gb']}G" if r["vram_used_gb"] is not None else "-",
+                  f"{r['power_w']:.0f}W" if r["power_w"] is not None else "-",
+                  f"{r['temp_c']:.0f}C" if r["temp_c"] else "-",
+                  r["owner"] or "-")
+    f.io.print(t)
+
+
+@cli.command("stats")
+@click.option("--no-stream", is_flag=True, help="print once and exit")
+@click.option("--interval", type=float, default=1.0, show_default=True)
+@pass_factory
+def stats_cmd(ctx: Ctx, no_stream, interval):
+    """Live per-sandbox + per-GPU stats (replaces docker stats)."""
+    f = ctx.factory
+    from ..monitor.stats import collect_stats, render_stats
+    if no_stream or not f.io.is_stdout_tty():
+        snap = collect_stats(f.engine())
+        f.io.print(render_stats(snap))
+        return
+    from rich.live import Live
+    with Live(console=f.io.console, refresh_per_second=4) as live:
+        while True:
+            snap = collect_stats(f.engine())
+            live.update(render_stats(snap))
+            time.sleep(interval)
+
+
+@cli.group("monitor")
+def monitor_group():
+    """Telemetry exporter + dashboards."""
+
+
+@monitor_group.command("up")
+@pass_factory
+def monitor_up(ctx: Ctx):
+    """Start the Prometheus metrics exporter daemon."""
+    from ..monitor.exporter import ensure_running
+    port = ctx.factory.config().settings.monitoring.prometheus_port
+    ensure_running(port)
+    ctx.factory.io.success(f"metrics exporter on :{port}/metrics")
+
+
+@monitor_group.command("down")
+@pass_factory
+def monitor_down(ctx: Ctx):
+    from ..monitor.exporter import stop_running
+    if stop_running():
+        ctx.factory.io.success("metrics exporter stopped")
+    else:
+        ctx.factory.io.eprint("metrics exporter not running")
+
+
+@monitor_group.command("status")
+@pass_factory
+def monitor_status(ctx: Ctx):
+    from ..monitor.exporter import exporter_running
+    port = ctx.factory.config().settings.monitoring.prometheus_port
+    ctx.factory.io.print(json.dumps(
+        {"running": exporter_running(), "port": port}))
+
+
+@cli.command("dashboard")
+@click.option("--interval", type=float, default=1.0, show_default=True)
+@pass_factory
+def dashboard_cmd(ctx: Ctx, interval):
+    """Full-screen live dashboard: agents x GPUs (reference: tui
+    RunDashboard precedent)."""
+    f = ctx.factory
+    from ..monitor.stats import collect_stats
+    from ..tui.dashboard import render_dashboard
+    from rich.live import Live
+    with Live(console=f.io.console, refresh_per_second=4, screen=True) as live:
+        try:
+            while True:
+                snap = collect_stats(f.engine())
+                live.update(render_dashboard(snap))
+                time.sleep(interval)
+        except KeyboardInterrupt:
+            pass

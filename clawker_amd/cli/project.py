@@ -1,0 +1,144 @@
+"""Project verbs + `clawker init` (reference: internal/cmd/project +
+project init first-run UX)."""
+from __future__ import annotations
+
+import json
+from pathlib import Path
+
+import click
+import yaml
+
+from .. import consts
+from ..config.config import slugify
+from ..errors import ClawkerError
+from ..project import ProjectRegistry
+from .root import Ctx, cli, pass_factory
+
+PRESETS = {
+    "python": {"build": {"stacks": ["python"]}},
+    "node": {"build": {"stacks": ["node"]}},
+    "go": {"build": {"stacks": ["go"]}},
+    "rust": {"build": {"stacks": ["rust"]}},
+    "cpp": {"build": {"stacks": ["cpp"]}},
+    "rocm": {"build": {"stacks": ["rocm", "python"]}, "gpu": {"count": 1}},
+}
+
+VCS_EGRESS = {
+    "github": ["github.com", "api.github.com", "codeload.github.com",
+               "raw.githubusercontent.com", "objects.githubusercontent.com"],
+    "gitlab": ["gitlab.com", "registry.gitlab.com"],
+    "bitbucket": ["bitbucket.org", "api.bitbucket.org"],
+}
+
+
+@cli.command("init")
+@click.option("--name", default="", help="project slug (default: directory name)")
+@click.option("--preset", type=click.Choice(sorted(PRESETS)), default=None)
+@click.option("--harness", default="claude", show_default=True)
+@click.option("--gpus", type=int, default=0, help="GPUs per agent sandbox")
+@click.option("--vcs", type=click.Choice(sorted(VCS_EGRESS)), multiple=True,
+              help="merge VCS egress domains")
+@click.option("-y", "--yes", is_flag=True, help="non-interactive")
+@pass_factory
+def init_cmd(ctx: Ctx, name, preset, harness, gpus, vcs, yes):
+    """Initialize a clawker project in the current directory."""
+    f = ctx.factory
+    root = f.cwd.resolve()
+    cfg_path = root / consts.PROJECT_FILE_NAME
+    if cfg_path.exists() or (root / consts.PROJECT_DIR_NAME).exists():
+        raise ClawkerError(f"project already initialized at {root}")
+    slug = slugify(name or root.name)
+
+    if not yes and f.io.can_prompt():
+        p = f.prompter()
+        slug = slugify(p.string("project name", slug))
+        harness = p.string("agent harness (claude/codex/echo)", harness)
+        if preset is None:
+            sel = p.select("language preset", ["none"] + sorted(PRESETS), 0)
+            preset = None if sel == "none" else sel
+
+    doc: dict = {"project": slug, "agent": {"harness": harness}}
+    if preset:
+        for k, v in PRESETS[preset].items():
+            doc.setdefault(k, {}).update(v)
+    if gpus:
+        doc.setdefault("gpu", {})["count"] = gpus
+    add_domains = []
+    for v in vcs:
+        add_domains += VCS_EGRESS[v]
+    if add_domains:
+        doc.setdefault("security", {})["egress"] = [
+            {"dst": d, "proto": "tls", "port": 443} for d in add_domains]
+    cfg_path.write_text(yaml.safe_dump(doc, sort_keys=False))
+    ignore = root / consts.IGNORE_FILE_NAME
+    if not ignore.exists():
+        ignore.write_text("# paths excluded from snapshot workspaces\n.git/\n")
+    ProjectRegistry().register(slug, root)
+    f.io.success(f"initialized project '{slug}' ({cfg_path.name}); "
+                 f"run `clawker build` then `clawker run -it`")
+
+
+@cli.group("project")
+def project_group():
+    """Manage registered projects."""
+
+
+@project_group.command("list")
+@click.option("--format", "fmt", default="")
+@pass_factory
+def project_list(ctx: Ctx, fmt):
+    f = ctx.factory
+    reg = ProjectRegistry()
+    entries = reg.list_projects()
+    # enrich with live sandbox counts (reference: manager.go ProjectState)
+    infos = f.engine().list()
+    if fmt == "json":
+        f.io.print(json.dumps([{
+            **p.__dict__,
+            "sandboxes": sum(1 for i in infos if i.project == p.name),
+            "running": sum(1 for i in infos if i.project == p.name and i.state == "running"),
+        } for p in entries], indent=1))
+        return
+    from rich.table import Table
+    t = Table(box=None, pad_edge=False)
+    for c in ("NAME", "ROOT", "SANDBOXES", "RUNNING"):
+        t.add_column(c)
+    for p in entries:
+        n_all = sum(1 for i in infos if i.project == p.name)
+        n_run = sum(1 for i in infos if i.project == p.name and i.state == "running")
+        t.add_row(p.name, p.root, str(n_all), str(n_run))
+    f.io.print(t)
+
+
+@project_group.command("register")
+@click.option("--name", default="")
+@pass_factory
+def project_register(ctx: Ctx, name):
+    """Register the current project directory in the registry."""
+    f = ctx.factory
+    cfg = f.config(require_project=True)
+    root = cfg.project_root
+    assert root is not None
+    entry = ProjectRegistry().register(name or cfg.project_slug, root)
+    f.io.success(f"registered '{entry.name}' at {entry.root}")
+
+
+@project_group.command("info")
+@pass_factory
+def project_info(ctx: Ctx):
+    f = ctx.factory
+    cfg = f.config(require_project=True)
+    from ..storage.store import to_plain
+    f.io.print(json.dumps({
+        "project": cfg.project_slug,
+        "root": str(cfg.project_root),
+        "config": to_plain(cfg.project),
+    }, indent=1))
+
+
+@project_group.command("remove")
+@click.argument("name")
+@pass_factory
+def project_remove(ctx: Ctx, name):
+    ProjectRegistry().unregister(name)
+    ctx.factory.io.success(f"unregistered '{name}'")

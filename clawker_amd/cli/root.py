@@ -1,0 +1,77 @@
+"""clawker CLI root (reference: internal/cmd/root — Cobra root with 20+
+command groups plus Docker-style top-level aliases; here click).
+
+Layout mirrors the reference's verb groups:
+  clawker run/create/start/stop/ps/rm/exec/logs/attach/wait/stats/kill  (container aliases)
+  clawker container <verb>      (the full group)
+  clawker image build/ls/rm/inspect/prune
+  clawker volume ls/rm
+  clawker project init/list/info/remove ; clawker init
+  clawker worktree add/list/remove/prune
+  clawker firewall status/list/add/remove/enable/disable/bypass/reload
+  clawker cp up/down/status/agents      (control plane)
+  clawker monitor up/down/status ; clawker gpus
+  clawker settings edit/get/set ; clawker alias
+  clawker version
+"""
+from __future__ import annotations
+
+import sys
+
+import click
+
+from .. import __version__, consts, logger
+from ..cmdutil import Factory
+from ..errors import ClawkerError, FlagError, SilentError
+
+
+class Ctx:
+    def __init__(self):
+        self.factory = Factory()
+
+
+pass_factory = click.make_pass_decorator(Ctx, ensure=True)
+
+
+@click.group(context_settings={"help_option_names": ["-h", "--help"]})
+@click.version_option(__version__, prog_name="clawker")
+def cli():
+    """MI355X-native agent-in-container orchestrator.
+
+    Runs AI coding-agent harnesses in isolated sandboxes with per-agent
+    GPU pinning, deny-by-default egress and live ROCm telemetry."""
+    logger.setup(consts.log_dir() / "clawker.log")
+
+
+def main() -> int:
+    from . import container, firewall, image, monitor, project, settings, cp, worktree  # noqa
+    try:
+        cli(standalone_mode=False)
+        return 0
+    except click.exceptions.Abort:
+        return 130
+    except click.exceptions.ClickException as e:
+        e.show()
+        return e.exit_code
+    except FlagError as e:
+        click.echo(f"error: {e}", err=True)
+        return 2
+    except SilentError as e:
+        return e.exit_code
+    except ClawkerError as e:
+        msg = e.user_message()
+        if msg:
+            click.echo(f"clawker: {msg}", err=True)
+        return e.exit_code
+    except KeyboardInterrupt:
+        return 130
+
+
+# import groups at module load so `clawker --help` lists them
+from . import container, firewall, image, monitor, project, settings, cp, worktree  # noqa: E402,F401
+
+if __name__ == "__main__":
+    # re-enter through the canonical module path so command registration
+    # (which imports clawker_amd.cli.root) targets THIS cli object
+    from clawker_amd.cli.root import main as _main
+    sys.exit(_main())

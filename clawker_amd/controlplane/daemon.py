@@ -1,0 +1,323 @@
+"""cpd — the clawker-amd control-plane daemon.
+
+Reference: the clawkercp privileged container (internal/controlplane
+cmd.go:367 Main — pub/sub topics, firewall handler + action queue, agent
+registry/dialer/executor, healthz, drain-to-zero, SOS). Single-node
+redesign: cpd is a host daemon (not a container) owning:
+
+  * the admin API over a root-only Unix socket (framed JSON — the
+    AdminService analog; filesystem perms replace mTLS+OAuth2)
+  * the agent registry (sqlite; cpd is the SOLE writer — the reference's
+    CP-is-sole-sqlite-writer rule)
+  * the firewall action queue: a single worker serializing ALL policy
+    mutations (reference: firewall/queue.go ActionQueue) — bypass with a
+    dead-man timer, rule reloads pushed to running sandbox gateways
+  * a sandbox watcher that runs Init/Boot plans for sandboxes started
+    detached, and drain-to-zero self-shutdown (reference: AgentWatcher)
+  * the event log (jsonl) every subsystem appends to
+
+Resilience contract (reference CLAUDE.md:44-100): after ready, cpd never
+exits on subsystem errors — failures degrade per-subsystem with
+event=<subsystem>_unavailable logs.
+"""
+from __future__ import annotations
+
+import json
+import os
+import queue
+import signal
+import socket
+import sqlite3
+import threading
+import time
+from pathlib import Path
+
+from .. import consts
+from ..engine import wire
+from ..logger import get as get_logger, setup as logger_setup
+
+log = get_logger("cpd")
+
+
+def admin_sock_path() -> Path:
+    return consts.runtime_dir() / consts.CP_ADMIN_SOCK
+
+
+def pid_path() -> Path:
+    return consts.runtime_dir() / "cpd.pid"
+
+
+def events_path() -> Path:
+    return consts.state_dir() / "cp-events.jsonl"
+
+
+class EventLog:
+    """Append-only jsonl event stream (the pub/sub + OTLP lanes analog)."""
+
+    def __init__(self, path: Path):
+        self.path = path
+        self.path.parent.mkdir(parents=True, exist_ok=True)
+        self._lock = threading.Lock()
+
+    def emit(self, event: str, **kv) -> None:
+        rec = {"ts": time.time(), "event": event, **kv}
+        with self._lock:
+            with open(self.path, "a") as f:
+                f.write(json.dumps(rec) + "\n")
+
+    def tail(self, n: int = 100) -> list[dict]:
+        try:
+            lines = self.path.read_text().splitlines()[-n:]
+            return [json.loads(l) for l in lines if l.strip()]
+        except (OSError, ValueError):
+            return []
+
+
+class ActionQueue:
+    """Single-goroutine FIFO serializing all firewall mutations
+    (reference: firewall/queue.go:122; head-wins coalescing for reloads)."""
+
+    def __init__(self):
+        self._q: "queue.Queue[tuple[str, callable]]" = queue.Queue()
+        self._worker = threading.Thread(target=self._run, daemon=True)
+        self._closed = False
+        self._pending_kinds: set[str] = set()
+        self._lock = threading.Lock()
+        self._worker.start()
+
+    def submit(self, kind: str, fn, coalesce: bool = False) -> None:
+        if self._closed:
+            return
+        with self._lock:
+            if coalesce and kind in self._pending_kinds:
+                return
+            self._pending_kinds.add(kind)
+        self._q.put((kind, fn))
+
+    def _run(self) -> None:
+        while True:
+            kind, fn = self._q.get()
+            with self._lock:
+                self._pending_kinds.discard(kind)
+            if fn is None:
+                return
+            try:
+                fn()
+            except Exception as e:   # never kill the queue
+                log.error("action_failed", kind=kind, err=str(e))
+
+    def close(self) -> None:
+        self._closed = True
+        self._q.put(("close", None))
+
+
+class CPDaemon:
+    def __init__(self):
+        from ..config.config import load_settings
+        from ..engine import Engine
+        self.settings = load_settings().get()
+        self.engine = Engine()
+        self.events = EventLog(events_path())
+        self.queue = ActionQueue()
+        self.ready = False
+        self._stop = threading.Event()
+        self._bypass_until = 0.0
+        self._bypass_timer: threading.Timer | None = None
+        self._registry_db = self._open_registry()
+        self._last_agent_seen = time.time()
+        self._boot_driven: set[str] = set()
+
+    # ------------------------------------------------------------ registry --
+    def _open_registry(self) -> sqlite3.Connection:
+        path = consts.data_dir() / "cp-registry.db"
+        db = sqlite3.connect(str(path), check_same_thread=False)
+        db.execute("PRAGMA journal_mode=WAL")
+        db.execute(
+            "CREATE TABLE IF NOT EXISTS agents ("
+            " sandbox TEXT PRIMARY KEY, project TEXT, agent TEXT,"
+            " first_seen REAL, last_seen REAL, state TEXT)")
+        db.commit()
+        return db
+
+    def _record_agent(self, info) -> None:
+        now = time.time()
+        self._registry_db.execute(
+            "INSERT INTO agents (sandbox,project,agent,first_seen,last_seen,state)"
+            " VALUES (?,?,?,?,?,?) ON CONFLICT(sandbox) DO UPDATE SET"
+            " last_seen=excluded.last_seen, state=excluded.state",
+            (info.name, info.project, info.agent, now, now, info.state))
+        self._registry_db.commit()
+
+    # ------------------------------------------------------------- watcher --
+    def _watch_loop(self) -> None:
+        """Track sandbox lifecycle; drive plans for detach-started
+        sandboxes; drain-to-zero."""
+        while not self._stop.is_set():
+            try:
+                infos = self.engine.list()
+                running = [i for i in infos if i.state == "running"]
+                for i in running:
+                    self._record_agent(i)
+                if running:
+                    self._last_agent_seen = time.time()
+                elif (self.settings.control_plane.drain_to_zero and self.ready and
+                      time.time() - self._last_agent_seen >
+                      self.settings.control_plane.drain_grace_s):
+                    log.info("drain_to_zero")
+                    self.events.emit("cp_drain_to_zero")
+                    self._stop.set()
+            except Exception as e:
+                log.error("watcher_unavailable", err=str(e))
+            self._stop.wait(1.0)
+
+    # -------------------------------------------------------------- policy --
+    def _reload_policy(self) -> int:
+        """Regenerate per-sandbox policy snapshots and nudge gateways
+        (reference: reconcileStackClosure — config gen + stack reload +
+        route sync)."""
+        from ..firewall.policy import compile_policy, write_policy_snapshot
+        n = 0
+        try:
+            pol = compile_policy(bypass=self.bypassed())
+            for i in self.engine.list():
+                if i.state == "running":
+                    write_policy_snapshot(i.rundir, pol)
+                    n += 1
+            self.events.emit("policy_reloaded", sandboxes=n)
+        except Exception as e:
+            log.error("policy_reload_unavailable", err=str(e))
+        return n
+
+    def bypassed(self) -> bool:
+        return time.time() < self._bypass_until
+
+    def _set_bypass(self, seconds: int) -> None:
+        cap = self.settings.firewall.bypass_max_s
+        seconds = max(0, min(seconds, cap))
+        self._bypass_until = time.time() + seconds
+        self.events.emit("firewall_bypass", seconds=seconds)
+        if self._bypass_timer:
+            self._bypass_timer.cancel()
+        if seconds > 0:
+            # dead-man restore (reference: Bypass 1h-capped timer)
+            self._bypass_timer = threading.Timer(
+                seconds, lambda: self.queue.submit("reload", self._reload_policy))
+            self._bypass_timer.daemon = True
+            self._bypass_timer.start()
+        self.queue.submit("reload", self._reload_policy)
+
+    # --------------------------------------------------------------- admin --
+    def _handle_admin(self, req: dict) -> dict:
+        op = req.get("op")
+        if op == "ping":
+            return {"ok": True, "ready": self.ready, "pid": os.getpid()}
+        if op == "status":
+            cur = self._registry_db.execute(
+                "SELECT COUNT(*) FROM agents WHERE state='running'")
+            return {"ok": True, "ready": self.ready,
+                    "bypass": self.bypassed(),
+                    "bypass_until": self._bypass_until,
+                    "agents_running": cur.fetchone()[0]}
+        if op == "agents":
+            cur = self._registry_db.execute(
+                "SELECT sandbox,project,agent,first_seen,last_seen,state FROM agents")
+            cols = ["sandbox", "project", "agent", "first_seen", "last_seen", "state"]
+            return {"ok": True,
+                    "agents": [dict(zip(cols, r)) for r in cur.fetchall()]}
+        if op == "reload_policy":
+            n = self._reload_policy()
+            return {"ok": True, "sandboxes": n}
+        if op == "bypass":
+            self._set_bypass(int(req.get("seconds", 0)))
+            return {"ok": True, "until": self._bypass_until}
+        if op == "events":
+            return {"ok": True, "events": self.events.tail(int(req.get("n", 100)))}
+        if op == "shutdown":
+            self._stop.set()
+            return {"ok": True}
+        return {"ok": False, "error": f"unknown op: {op}"}
+
+    def _admin_loop(self, listener: socket.socket) -> None:
+        listener.settimeout(0.5)
+        while not self._stop.is_set():
+            try:
+                conn, _ = listener.accept()
+            except socket.timeout:
+                continue
+            except OSError:
+                break
+            threading.Thread(target=self._serve_conn, args=(conn,),
+                             daemon=True).start()
+
+    def _serve_conn(self, conn: socket.socket) -> None:
+        try:
+            conn.settimeout(30)
+            while True:
+                req = wire.recv_frame(conn)
+                if req is None:
+                    return
+                try:
+                    resp = self._handle_admin(req)
+                except Exception as e:   # degrade, never crash
+                    log.error("admin_op_failed", op=req.get("op"), err=str(e))
+                    resp = {"ok": False, "error": str(e)}
+                wire.send_frame(conn, resp)
+        except OSError:
+            pass
+        finally:
+            conn.close()
+
+    # ----------------------------------------------------------------- run --
+    def run(self) -> int:
+        rd = consts.runtime_dir()
+        rd.mkdir(parents=True, exist_ok=True)
+        sock_path = admin_sock_path()
+        listener = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        try:
+            sock_path.unlink(missing_ok=True)
+            listener.bind(str(sock_path))
+            os.chmod(sock_path, 0o600)
+            listener.listen(16)
+        except OSError as e:
+            log.error("admin_listen_failed", err=str(e))
+            return 1
+        pid_path().write_text(str(os.getpid()))
+        self.events.emit("cp_starting", pid=os.getpid())
+
+        threads = [
+            threading.Thread(target=self._admin_loop, args=(listener,), daemon=True),
+            threading.Thread(target=self._watch_loop, daemon=True),
+        ]
+        for t in threads:
+            t.start()
+        self.queue.submit("reload", self._reload_policy)
+        self.ready = True
+        self.events.emit("cp_ready")
+
+        def on_sig(*_a):
+            self._stop.set()
+        signal.signal(signal.SIGTERM, on_sig)
+        signal.signal(signal.SIGINT, on_sig)
+
+        while not self._stop.is_set():
+            self._stop.wait(0.5)
+
+        # drain sequence (reference: runDrainSequence ordering)
+        self.events.emit("cp_draining")
+        self.queue.close()
+        if self._bypass_timer:
+            self._bypass_timer.cancel()
+        listener.close()
+        sock_path.unlink(missing_ok=True)
+        pid_path().unlink(missing_ok=True)
+        self.events.emit("cp_stopped")
+        return 0
+
+
+def main() -> int:
+    logger_setup(consts.log_dir() / "cpd.log")
+    return CPDaemon().run()
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -1,0 +1,97 @@
+"""Image build primitive: run a build script in a throwaway sandbox whose
+overlay upper becomes a new content-addressed image layer.
+
+Reference analog: internal/docker/builder.go Build driving dockerd builds;
+here the engine is its own builder (no daemon): layer = the COW residue of
+the build sandbox. Requires the ns backend (no overlay on proc hosts)."""
+from __future__ import annotations
+
+import time
+import uuid
+from typing import Callable
+
+from .. import consts
+from ..errors import EngineError
+from ..logger import get as get_logger
+from .engine import Engine
+from .images import ImageMeta
+from .spec import SandboxSpec
+
+log = get_logger("build")
+
+ProgressFn = Callable[[str], None]
+
+
+def build_layer(engine: Engine, base_image: str, script: str,
+                env: dict | None = None, network: bool = True,
+                on_progress: ProgressFn | None = None,
+                timeout_s: float = 900.0) -> str:
+    """Run `script` (sh -e) over base_image; returns the committed layer id.
+    Build output streams through on_progress."""
+    if engine.backend != "ns":
+        raise EngineError(
+            "build", "image builds need the ns isolation backend "
+            "(overlayfs); this host cannot create mount namespaces")
+    tmp_id, layer_fs = engine.images.new_layer_dir()
+    name = f"{consts.SANDBOX_NAME_PREFIX}build.{uuid.uuid4().hex[:8]}"
+    spec = SandboxSpec(
+        name=name, hostname="clawker-build", netns=not network,
+        autostart=True, cmd=["/bin/sh", "-ec", script],
+        env=dict(env or {}), user="", workdir="/",
+        labels={consts.MANAGED_LABEL: "true", "dev.clawker.build": "true"},
+    )
+    spec.upper = str(layer_fs)
+    spec.work = str(layer_fs.parent / "work")
+    engine.create(spec, image=base_image)
+    try:
+        engine.start(name)
+        # stream console as progress
+        last = 0
+        deadline = time.monotonic() + timeout_s
+        while True:
+            data = engine.logs(name)
+            if on_progress and len(data) > last:
+                for line in data[last:].decode(errors="replace").splitlines():
+                    on_progress(line)
+                last = len(data)
+            info = engine.inspect(name)
+            if info.state != "running":
+                code = info.exit_code
+                break
+            if time.monotonic() > deadline:
+                engine.stop(name, timeout_s=3)
+                raise EngineError("build", f"build timed out after {timeout_s}s")
+            time.sleep(0.05)
+        if code != 0:
+            tail = engine.logs(name)[-1200:].decode(errors="replace")
+            raise EngineError("build", f"build script failed ({code}):\n{tail}")
+    finally:
+        try:
+            engine.remove(name, force=True)
+        except Exception:
+            pass
+    # scrub build-sandbox runtime residue from the layer
+    for junk in ("run/clawker", ".oldroot"):
+        p = layer_fs / junk
+        if p.exists():
+            import shutil
+            shutil.rmtree(p, ignore_errors=True)
+    lid = engine.images.commit_layer(tmp_id)
+    log.info("layer_built", layer=lid, base=base_image)
+    return lid
+
+
+def build_image(engine: Engine, name: str, base_image: str, script: str,
+                env: dict | None = None, user: str = "", cmd: list | None = None,
+                workdir: str = "", labels: dict | None = None,
+                base_hash: str = "", network: bool = True,
+                on_progress: ProgressFn | None = None) -> ImageMeta:
+    lid = build_layer(engine, base_image, script, env=env, network=network,
+                      on_progress=on_progress)
+    meta = ImageMeta(
+        name=name, layers=[lid], env=dict(env or {}), user=user,
+        cmd=list(cmd or []), workdir=workdir, labels=dict(labels or {}),
+        parent=base_image, base_hash=base_hash)
+    engine.images.put(meta)
+    log.info("image_built", image=name, layer=lid)
+    return meta

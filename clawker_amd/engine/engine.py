@@ -184,8 +184,10 @@ class Engine:
         spec.rundir = str(rundir)
         if self.backend == "ns":
             spec.lowerdirs = self.images.lowerdirs_for(image)
-            spec.upper = str(statedir / "upper")
-            spec.work = str(statedir / "work")
+            if not spec.upper:        # builds set a staging layer as upper
+                spec.upper = str(statedir / "upper")
+            if not spec.work:         # workdir must share the upper's fs
+                spec.work = str(statedir / "work")
             spec.merged = str(rundir / "merged")
             if self._passthrough is None:
                 self._passthrough = host_passthrough_mounts()

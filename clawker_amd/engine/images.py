@@ -62,6 +62,33 @@ class ImageStore:
         self.root = root or consts.image_store_dir()
         (self.root / "layers").mkdir(parents=True, exist_ok=True)
         (self.root / "meta").mkdir(parents=True, exist_ok=True)
+        self._layerfs_checked = False
+
+    def ensure_layer_filesystem(self) -> None:
+        """Overlayfs forbids a lowerdir nested inside another lowerdir
+        (mount-time ELOOP trap), and our bottom lower is the host "/" — so
+        the layer store must live on its OWN superblock. If images/layers
+        shares st_dev with /, mount a tmpfs over it (host-wide, survives
+        the process; a production node should dedicate a real partition or
+        bind a non-root filesystem at CLAWKER_IMAGE_DIR instead)."""
+        if self._layerfs_checked:
+            return
+        self._layerfs_checked = True
+        layers = self.root / "layers"
+        try:
+            if os.stat(layers).st_dev != os.stat("/").st_dev:
+                return   # already separate (dedicated partition / prior mount)
+            if os.geteuid() != 0:
+                return
+            import subprocess
+            r = subprocess.run(
+                ["mount", "-t", "tmpfs", "-o", "mode=700", "clawker-layers",
+                 str(layers)], capture_output=True, text=True)
+            if r.returncode != 0:
+                from ..logger import get as _get
+                _get("images").warn("layerfs_unavailable", err=r.stderr.strip())
+        except OSError:
+            pass
 
     # -- metadata ------------------------------------------------------------
     def _meta_path(self, name: str) -> Path:
@@ -105,15 +132,19 @@ class ImageStore:
     # -- layers ----------------------------------------------------------------
     def new_layer_dir(self) -> tuple[str, Path]:
         """Allocate a staging layer; caller fills fs/ then commit_layer()."""
+        self.ensure_layer_filesystem()
         lid = "tmp-" + uuid.uuid4().hex[:12]
         d = self.root / "layers" / lid / "fs"
         d.mkdir(parents=True)
+        # overlay workdir must share the upper's filesystem
+        (self.root / "layers" / lid / "work").mkdir()
         return lid, d
 
     def commit_layer(self, tmp_id: str) -> str:
         """Rename a staging layer to its content id (cheap pseudo-hash:
         file list + sizes + mtimes digest — enough for staleness checks)."""
         src = self.root / "layers" / tmp_id
+        shutil.rmtree(src / "work", ignore_errors=True)
         h = hashlib.sha256()
         fs = src / "fs"
         for p in sorted(fs.rglob("*")):

@@ -1,0 +1,94 @@
+"""Stats collection: per-sandbox cgroup metrics + per-GPU rocm telemetry
+(replaces the reference's docker-stats stream, stats.go:194)."""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from pathlib import Path
+
+from ..engine import Engine
+from .sampler import GpuSample, RocmSampler
+
+_CG_ROOT = Path("/sys/fs/cgroup")
+
+
+def _read_int(p: Path) -> int | None:
+    try:
+        v = p.read_text().strip()
+        return int(v) if v.isdigit() else None
+    except OSError:
+        return None
+
+
+def cgroup_stats(name: str) -> dict:
+    """Memory + pids for a sandbox's cgroup (v1 hybrid or v2 unified)."""
+    out: dict = {}
+    v2 = _CG_ROOT / "clawker" / name
+    if (v2 / "memory.current").exists():
+        out["mem_bytes"] = _read_int(v2 / "memory.current")
+        out["pids"] = _read_int(v2 / "pids.current")
+        return out
+    mem = _CG_ROOT / "memory" / "clawker" / name / "memory.usage_in_bytes"
+    pids = _CG_ROOT / "pids" / "clawker" / name / "pids.current"
+    out["mem_bytes"] = _read_int(mem)
+    out["pids"] = _read_int(pids)
+    return out
+
+
+@dataclass
+class StatsSnapshot:
+    ts: float = field(default_factory=time.time)
+    sandboxes: list[dict] = field(default_factory=list)
+    gpus: list[GpuSample] = field(default_factory=list)
+    allocations: dict = field(default_factory=dict)   # gpu index -> sandbox
+
+
+_sampler: RocmSampler | None = None
+
+
+def collect_stats(engine: Engine) -> StatsSnapshot:
+    global _sampler
+    snap = StatsSnapshot()
+    for info in engine.list():
+        row = {
+            "name": info.name, "state": info.state, "pid": info.pid,
+            "gpus": info.gpus, "agent": info.agent, "project": info.project,
+        }
+        if info.state == "running":
+            row.update(cgroup_stats(info.name))
+        snap.sandboxes.append(row)
+        for g in info.gpus:
+            snap.allocations[g] = info.name
+    try:
+        if _sampler is None:
+            _sampler = RocmSampler()
+        snap.gpus = _sampler.sample()
+    except Exception:
+        snap.gpus = []
+    return snap
+
+
+def render_stats(snap: StatsSnapshot):
+    from rich.console import Group
+    from rich.table import Table
+    t = Table(title="sandboxes", box=None, pad_edge=False)
+    for c in ("NAME", "STATE", "PID", "MEM", "PIDS", "GPUS"):
+        t.add_column(c)
+    for s in snap.sandboxes:
+        mem = s.get("mem_bytes")
+        t.add_row(s["name"], s["state"], str(s.get("pid") or "-"),
+                  f"{mem / 2**20:.0f}M" if mem else "-",
+                  str(s.get("pids") or "-"),
+                  ",".join(map(str, s["gpus"])) or "-")
+    if not snap.gpus:
+        return t
+    g = Table(title="GPUs (MI355X)", box=None, pad_edge=False)
+    for c in ("GPU", "BUSY", "VRAM", "POWER", "TEMP(J)", "SCLK", "OWNER"):
+        g.add_column(c)
+    for s in snap.gpus:
+        g.add_row(str(s.index), f"{s.busy_pct:.0f}%",
+                  f"{s.vram_used / 2**30:.1f}/{s.vram_total / 2**30:.0f}G",
+                  f"{s.power_w:.0f}W", f"{s.temp_junction_c:.0f}C",
+                  f"{s.sclk_mhz:.0f}MHz",
+                  snap.allocations.get(s.index, "-"))
+    return Group(t, g)

@@ -1,0 +1,112 @@
+"""Image build pipeline tests (ns backend: overlay layer builds)."""
+import json
+from pathlib import Path
+
+import pytest
+
+from conftest import requires_isolation
+
+
+@pytest.fixture
+def ctx(isolated_env, tmp_path):
+    root = tmp_path / "bproj"
+    root.mkdir()
+    (root / ".clawker.yaml").write_text(
+        "project: btest\nagent:\n  harness: echo\n"
+        "build:\n  steps:\n    - echo project-step > /etc/project-step\n")
+    from clawker_amd.config import load_config
+    from clawker_amd.engine import Engine
+    cfg = load_config(root)
+    eng = Engine()
+    yield cfg, eng
+    for info in eng.list():
+        try:
+            eng.remove(info.name, force=True)
+        except Exception:
+            pass
+    eng.close()
+
+
+@requires_isolation
+def test_build_layer_commits_cow_residue(ctx):
+    cfg, eng = ctx
+    from clawker_amd.engine.build import build_layer
+    lid = build_layer(eng, "hostfs", "echo hi > /layer-marker\nmkdir -p /opt/x")
+    fs = eng.images.layer_path(lid)
+    assert (fs / "layer-marker").read_text().strip() == "hi"
+    assert (fs / "opt/x").is_dir()
+    # build sandbox runtime residue scrubbed
+    assert not (fs / "run/clawker").exists()
+
+
+@requires_isolation
+def test_build_failure_reports_script_tail(ctx):
+    cfg, eng = ctx
+    from clawker_amd.engine.build import build_layer
+    from clawker_amd.errors import EngineError
+    with pytest.raises(EngineError, match="boom-marker"):
+        build_layer(eng, "hostfs", "echo boom-marker; exit 3")
+
+
+@requires_isolation
+def test_two_stage_project_build_and_run(ctx):
+    cfg, eng = ctx
+    from clawker_amd.bundler import Builder
+    lines = []
+    name = Builder(cfg, eng).build(on_progress=lines.append)
+    assert name == "clawker-btest:echo"
+    assert eng.images.exists("clawker-btest:base")
+    assert eng.images.exists("clawker-btest:default")
+    meta = eng.images.get(name)
+    assert meta.parent == "clawker-btest:base"
+    assert meta.cmd[0] == "/bin/sh"          # echo harness CMD
+
+    # rebuild: base is cached (hash unchanged)
+    lines2 = []
+    Builder(cfg, eng).build(on_progress=lines2.append)
+    assert any("up to date" in l for l in lines2)
+
+    # run a sandbox FROM the image: project step + plan scripts visible
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    orch = Orchestrator(cfg, eng)
+    sb = "clawker.btest.fromimg"
+    orch.run(RunOptions(
+        agent="fromimg", name=sb, image=name, autostart=True, firewall=False,
+        cmd=["/bin/sh", "-c",
+             "cat /etc/project-step; ls /etc/clawker/; cat /etc/clawker/egress-floor.yaml | head -2"]))
+    assert eng.wait(sb, timeout_s=30) == 0
+    out = eng.logs(sb).decode()
+    assert "project-step" in out
+    assert "pre-run.sh" in out and "post-init.sh" in out
+    assert "harness: echo" in out
+    orch.teardown(sb, force=True)
+
+
+@requires_isolation
+def test_boot_plans_execute_from_image(ctx):
+    """InitPlan runs once (marker), BootPlan every start (reference:
+    init_steps.go / boot_steps.go semantics)."""
+    cfg, eng = ctx
+    from clawker_amd.bundler import Builder
+    from clawker_amd.cmdutil import Factory
+    from clawker_amd.controlplane.plans import run_boot_plans
+    from clawker_amd.iostreams import TestIOStreams
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    img = Builder(cfg, eng).build()
+    orch = Orchestrator(cfg, eng)
+    sb = "clawker.btest.plans"
+    orch.run(RunOptions(agent="plans", name=sb, image=img, autostart=False,
+                        firewall=False,
+                        cmd=["/bin/sh", "-c", "cat /tmp/clawker-hooks.log"]))
+    f = Factory(io=TestIOStreams())
+    with orch.client(sb) as c:
+        hello = c.hello()
+        assert hello["initialized"] is False
+        run_boot_plans(f, sb, c, hello, quiet=True)
+        c.agent_ready()
+    assert eng.wait(sb, timeout_s=30) == 0
+    out = eng.logs(sb).decode()
+    # echo harness hooks wrote into /tmp inside the sandbox
+    assert "post-init hook executed" in out
+    assert "pre-run hook executed" in out
+    orch.teardown(sb, force=True)

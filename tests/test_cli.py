@@ -1,0 +1,158 @@
+"""CLI tests through click's CliRunner + subprocess e2e (reference test
+strategy: command tests through the Factory seam; e2e through the real
+binary — SURVEY.md §4)."""
+import json
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+from click.testing import CliRunner
+
+from conftest import requires_isolation
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+@pytest.fixture
+def proj(isolated_env, tmp_path, monkeypatch):
+    root = tmp_path / "cliproj"
+    root.mkdir()
+    monkeypatch.chdir(root)
+    return root
+
+
+def _invoke(args, cwd=None):
+    """Invoke through click; ClawkerErrors map to exit codes like main()."""
+    from clawker_amd.cli.root import cli
+    from clawker_amd.errors import ClawkerError
+    runner = CliRunner()
+    r = runner.invoke(cli, args, catch_exceptions=True)
+    if r.exception is not None and not isinstance(r.exception, SystemExit):
+        if isinstance(r.exception, ClawkerError):
+            r.exit_code = r.exception.exit_code
+        else:
+            raise r.exception
+    return r
+
+
+def test_init_creates_config_and_registers(proj):
+    r = _invoke(["init", "--yes", "--name", "My Proj", "--harness", "echo",
+                 "--gpus", "2", "--vcs", "github"])
+    assert r.exit_code == 0, r.output
+    import yaml
+    doc = yaml.safe_load((proj / ".clawker.yaml").read_text())
+    assert doc["project"] == "my-proj"
+    assert doc["agent"]["harness"] == "echo"
+    assert doc["gpu"]["count"] == 2
+    dsts = [e["dst"] for e in doc["security"]["egress"]]
+    assert "github.com" in dsts
+    from clawker_amd.project import ProjectRegistry
+    assert ProjectRegistry().get("my-proj").root == str(proj)
+    # re-init refuses
+    r2 = _invoke(["init", "--yes"])
+    assert r2.exit_code != 0
+
+
+def test_version_and_help(isolated_env):
+    r = _invoke(["version"])
+    assert r.exit_code == 0 and "clawker-amd" in r.output
+    r = _invoke(["--help"])
+    assert "run" in r.output and "firewall" in r.output
+
+
+def test_firewall_rules_cli(proj):
+    _invoke(["init", "--yes", "--harness", "echo"])
+    r = _invoke(["firewall", "add", "api.example.com", "--port", "443",
+                 "--deny-path", "/share"])
+    assert r.exit_code == 0, r.output
+    r = _invoke(["firewall", "list", "--format", "json"])
+    rules = json.loads(r.output)
+    assert rules[0]["dst"] == "api.example.com"
+    assert rules[0]["deny_paths"] == ["/share"]
+    # dedupe with path merge
+    _invoke(["firewall", "add", "api.example.com", "--deny-path", "/public"])
+    rules = json.loads(_invoke(["firewall", "list", "--format", "json"]).output)
+    assert len(rules) == 1
+    assert set(rules[0]["deny_paths"]) == {"/share", "/public"}
+    r = _invoke(["firewall", "remove", "api.example.com"])
+    assert r.exit_code == 0
+    assert json.loads(_invoke(["firewall", "list", "--format", "json"]).output) == []
+
+
+def test_settings_get_set(isolated_env):
+    r = _invoke(["settings", "get", "firewall.bypass_max_s"])
+    assert json.loads(r.output) == 3600
+    r = _invoke(["settings", "set", "firewall.bypass_max_s", "1800"])
+    assert r.exit_code == 0
+    assert json.loads(_invoke(["settings", "get", "firewall.bypass_max_s"]).output) == 1800
+
+
+def test_alias_crud(isolated_env):
+    assert _invoke(["alias", "set", "r8", "run --gpus 8"]).exit_code == 0
+    assert "r8 = run --gpus 8" in _invoke(["alias", "list"]).output
+    assert _invoke(["alias", "delete", "r8"]).exit_code == 0
+    assert _invoke(["alias", "delete", "r8"]).exit_code != 0
+
+
+@requires_isolation
+def test_cli_run_ps_logs_rm_e2e(proj):
+    """Full verb flow against real sandboxes."""
+    _invoke(["init", "--yes", "--name", "e2e", "--harness", "echo"])
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+
+    def clawker(*args, timeout=60):
+        return subprocess.run([sys.executable, "-m", "clawker_amd", *args],
+                              capture_output=True, text=True, timeout=timeout,
+                              cwd=str(proj), env=env)
+
+    # detached run
+    r = clawker("run", "-d", "--agent", "w1", "--no-firewall",
+                "--", "/bin/sh", "-c", "echo from-w1; sleep 20")
+    assert r.returncode == 0, r.stderr
+    r = clawker("ps", "--format", "json")
+    rows = json.loads(r.stdout)
+    assert any(x["name"] == "clawker.e2e.w1" and x["state"] == "running" for x in rows)
+    # exec
+    r = clawker("exec", "w1", "--", "/bin/echo", "exec-ok")
+    assert r.returncode == 0 and "exec-ok" in r.stdout
+    # logs
+    r = clawker("logs", "w1")
+    assert "from-w1" in r.stdout
+    # stop + wait + rm
+    assert clawker("stop", "w1").returncode == 0
+    r = clawker("ps", "-a", "--format", "json")
+    assert any(x["name"] == "clawker.e2e.w1" and x["state"] == "exited"
+               for x in json.loads(r.stdout))
+    assert clawker("rm", "w1").returncode == 0
+    r = clawker("ps", "-a", "--format", "json")
+    assert not any(x["name"] == "clawker.e2e.w1" for x in json.loads(r.stdout))
+
+
+@requires_isolation
+def test_cli_run_exit_code_propagates(proj):
+    _invoke(["init", "--yes", "--name", "ec", "--harness", "echo"])
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    r = subprocess.run(
+        [sys.executable, "-m", "clawker_amd", "run", "--rm", "--no-firewall",
+         "--", "/bin/sh", "-c", "exit 9"],
+        capture_output=True, text=True, timeout=60, cwd=str(proj), env=env)
+    assert r.returncode == 9, (r.stdout, r.stderr)
+
+
+@requires_isolation
+def test_worktree_cli(proj):
+    _invoke(["init", "--yes", "--name", "wt", "--harness", "echo"])
+    subprocess.run(["git", "init", "-q", "-b", "main"], cwd=proj, check=True)
+    subprocess.run(["git", "-c", "user.email=t@t", "-c", "user.name=t",
+                    "commit", "-q", "--allow-empty", "-m", "init"],
+                   cwd=proj, check=True)
+    r = _invoke(["worktree", "add", "feature/x"])
+    assert r.exit_code == 0, r.output
+    r = _invoke(["worktree", "list", "--format", "json"])
+    rows = json.loads(r.output)
+    assert rows[0]["branch"] == "feature/x" and rows[0]["status"] == "ok"
+    r = _invoke(["worktree", "remove", "feature/x"])
+    assert r.exit_code == 0
+    assert json.loads(_invoke(["worktree", "list", "--format", "json"]).output) == []
