@@ -11,7 +11,11 @@ PY_INC := $(shell $(PY) -c "import sysconfig; print(sysconfig.get_paths()['inclu
 
 all: native pymod
 
-native: native/bin/ckrt native/bin/ckd
+native: native/bin/ckrt native/bin/ckd native/bin/ckgw
+
+native/bin/ckgw: native/ckgw/ckgw.cpp $(COMMON)
+	@mkdir -p native/bin
+	$(CXX) $(CXXFLAGS) -o $@ native/ckgw/ckgw.cpp
 
 pymod: clawker_amd/_native$(EXT_SUFFIX)
 

@@ -40,8 +40,8 @@ class CPClient:
         if self.running() and self._ping():
             return
         env = dict(os.environ)
-        logf = open(consts.log_dir() / "cpd.out", "ab")
         consts.log_dir().mkdir(parents=True, exist_ok=True)
+        logf = open(consts.log_dir() / "cpd.out", "ab")
         subprocess.Popen(
             [sys.executable, "-m", "clawker_amd.controlplane.daemon"],
             stdin=subprocess.DEVNULL, stdout=logf, stderr=logf,

@@ -115,17 +115,19 @@ class CPDaemon:
     def __init__(self):
         from ..config.config import load_settings
         from ..engine import Engine
+        from ..firewall.gateway import GatewayManager
         self.settings = load_settings().get()
         self.engine = Engine()
         self.events = EventLog(events_path())
         self.queue = ActionQueue()
+        self.gateways = GatewayManager(
+            on_event=lambda ev: self.events.emit("egress_decision", **ev))
         self.ready = False
         self._stop = threading.Event()
         self._bypass_until = 0.0
         self._bypass_timer: threading.Timer | None = None
         self._registry_db = self._open_registry()
         self._last_agent_seen = time.time()
-        self._boot_driven: set[str] = set()
 
     # ------------------------------------------------------------ registry --
     def _open_registry(self) -> sqlite3.Connection:
@@ -158,6 +160,16 @@ class CPDaemon:
                 running = [i for i in infos if i.state == "running"]
                 for i in running:
                     self._record_agent(i)
+                # reconcile firewall gateways against live state (reference:
+                # dockerevents reconcile + FirewallEnable drift guard)
+                live_fw = {i.name: i for i in running
+                           if i.labels.get("dev.clawker.fw") == "on"}
+                for name, i in live_fw.items():
+                    if name not in self.gateways.gateways:
+                        self._attach_gateway(name, i.rundir)
+                for name in list(self.gateways.gateways):
+                    if name not in live_fw:
+                        self.gateways.detach(name)
                 if running:
                     self._last_agent_seen = time.time()
                 elif (self.settings.control_plane.drain_to_zero and self.ready and
@@ -187,6 +199,14 @@ class CPDaemon:
         except Exception as e:
             log.error("policy_reload_unavailable", err=str(e))
         return n
+
+    def _attach_gateway(self, name: str, rundir) -> None:
+        from pathlib import Path as _P
+        from ..firewall.policy import compile_policy, write_policy_snapshot
+        rundir = _P(rundir)
+        write_policy_snapshot(rundir, compile_policy(bypass=self.bypassed()))
+        self.gateways.attach(name, rundir)
+        self.events.emit("firewall_enabled", sandbox=name)
 
     def bypassed(self) -> bool:
         return time.time() < self._bypass_until
@@ -230,6 +250,12 @@ class CPDaemon:
         if op == "bypass":
             self._set_bypass(int(req.get("seconds", 0)))
             return {"ok": True, "until": self._bypass_until}
+        if op == "fw_attach":
+            self._attach_gateway(req["sandbox"], req["rundir"])
+            return {"ok": True}
+        if op == "fw_detach":
+            self.gateways.detach(req["sandbox"])
+            return {"ok": True}
         if op == "events":
             return {"ok": True, "events": self.events.tail(int(req.get("n", 100)))}
         if op == "shutdown":
@@ -302,11 +328,13 @@ class CPDaemon:
         while not self._stop.is_set():
             self._stop.wait(0.5)
 
-        # drain sequence (reference: runDrainSequence ordering)
+        # drain sequence (reference: runDrainSequence ordering — queue
+        # close -> stop serving -> cancel bypass timers -> stack stop)
         self.events.emit("cp_draining")
         self.queue.close()
         if self._bypass_timer:
             self._bypass_timer.cancel()
+        self.gateways.detach_all()
         listener.close()
         sock_path.unlink(missing_ok=True)
         pid_path().unlink(missing_ok=True)

@@ -15,12 +15,9 @@ from . import wire
 class CkdClient:
     def __init__(self, sock_path: Path, timeout: float | None = 30.0):
         self.sock_path = sock_path
-        self.sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
-        self.sock.settimeout(timeout)
         try:
-            self.sock.connect(str(sock_path))
+            self.sock = wire.connect_unix(sock_path, timeout=timeout)
         except OSError as e:
-            self.sock.close()
             raise EngineError("ckd connect", f"{sock_path}: {e}") from e
 
     @classmethod

@@ -161,12 +161,16 @@ class Engine:
 
         # stage the PID-1 supervisor into the rundir (bind-mounted at
         # /run/clawker inside; ckrt execs /run/clawker/bin/ckd)
-        ckd_src = native_bin_dir() / "ckd"
-        if not ckd_src.is_file():
-            raise EngineError("create", f"ckd binary missing: {ckd_src} (run `make native`)")
-        ckd_dst = rundir / "bin" / "ckd"
-        if not ckd_dst.exists() or ckd_dst.stat().st_mtime < ckd_src.stat().st_mtime:
-            shutil.copy2(ckd_src, ckd_dst)
+        for binname in ("ckd", "ckgw"):
+            src = native_bin_dir() / binname
+            if not src.is_file():
+                if binname == "ckd":
+                    raise EngineError(
+                        "create", f"ckd binary missing: {src} (run `make native`)")
+                continue
+            dst = rundir / "bin" / binname
+            if not dst.exists() or dst.stat().st_mtime < src.stat().st_mtime:
+                shutil.copy2(src, dst)
 
         # per-sandbox identity files
         (rundir / "hostname").write_text(spec.hostname + "\n")

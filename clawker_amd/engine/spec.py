@@ -50,6 +50,8 @@ class SandboxSpec:
     # ckd path contract: where the control socket/console/ready live from
     # ckd's point of view, and where the one-time init marker persists
     paths: dict[str, str] = field(default_factory=dict)
+    # auxiliary in-sandbox daemons ckd supervises (e.g. ckgw gateway shims)
+    services: list[dict] = field(default_factory=list)
 
     def to_json(self) -> str:
         d = {
@@ -57,6 +59,7 @@ class SandboxSpec:
             "backend": self.backend,
             "rundir": self.rundir,
             "paths": self.paths,
+            "services": self.services,
             "rootfs": {
                 "lowerdirs": self.lowerdirs,
                 "upper": self.upper,

@@ -39,7 +39,10 @@ class StateDB:
     def __init__(self, path: Path | None = None):
         self.path = path or (consts.data_dir() / "state.db")
         self.path.parent.mkdir(parents=True, exist_ok=True)
-        self.db = sqlite3.connect(str(self.path), timeout=10.0)
+        # check_same_thread=False: the CP daemon reads from watcher/admin
+        # threads; python's sqlite3 serializes access internally
+        self.db = sqlite3.connect(str(self.path), timeout=10.0,
+                                  check_same_thread=False)
         self.db.execute("PRAGMA journal_mode=WAL")
         self.db.execute("PRAGMA busy_timeout=10000")
         self.db.executescript(_SCHEMA)

@@ -20,6 +20,40 @@ class WireError(Exception):
     pass
 
 
+def bind_unix(path, sock_type=socket.SOCK_STREAM) -> socket.socket:
+    """Bind a unix socket, dodging the 108-byte sun_path limit via an
+    O_PATH dirfd (/proc/self/fd/N/name)."""
+    import os
+    s = socket.socket(socket.AF_UNIX, sock_type)
+    p = str(path)
+    if len(p.encode()) < 100:
+        s.bind(p)
+        return s
+    fd = os.open(os.path.dirname(p), os.O_PATH)
+    try:
+        s.bind(f"/proc/self/fd/{fd}/{os.path.basename(p)}")
+    finally:
+        os.close(fd)
+    return s
+
+
+def connect_unix(path, timeout: float | None = None) -> socket.socket:
+    import os
+    s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    if timeout is not None:
+        s.settimeout(timeout)
+    p = str(path)
+    if len(p.encode()) < 100:
+        s.connect(p)
+        return s
+    fd = os.open(os.path.dirname(p), os.O_PATH)
+    try:
+        s.connect(f"/proc/self/fd/{fd}/{os.path.basename(p)}")
+    finally:
+        os.close(fd)
+    return s
+
+
 def send_frame(sock: socket.socket, obj: dict[str, Any]) -> None:
     body = json.dumps(obj, separators=(",", ":")).encode()
     if len(body) > MAX_FRAME:

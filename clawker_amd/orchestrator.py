@@ -162,12 +162,31 @@ class Orchestrator:
             consts.AGENT_LABEL: opts.agent,
             consts.HARNESS_LABEL: proj.agent.harness,
             consts.GPU_LABEL: ",".join(str(i) for i in gpu_indices),
+            "dev.clawker.fw": "on" if effective_firewall else "off",
         }
         labels.update(opts.labels)
+
+        # egress gateway shims: the only paths out of the uplink-less netns
+        # (see native/ckgw/ckgw.cpp; host side: firewall/gateway.py)
+        services: list[dict] = []
+        if effective_firewall:
+            services = [
+                {"name": "egress-gw",
+                 "argv": ["/run/clawker/bin/ckgw", "tcp", "127.0.0.1:3128",
+                          "/run/clawker/egress.sock"]},
+                {"name": "dns-gw",
+                 "argv": ["/run/clawker/bin/ckgw", "dns", "127.0.0.1:53",
+                          "/run/clawker/dns.sock"]},
+            ]
+            for var in ("HTTP_PROXY", "HTTPS_PROXY", "http_proxy", "https_proxy"):
+                env.setdefault(var, "http://127.0.0.1:3128")
+            env.setdefault("NO_PROXY", "localhost,127.0.0.1")
+            env.setdefault("no_proxy", "localhost,127.0.0.1")
 
         spec = SandboxSpec(
             name=name,
             hostname=f"{self.cfg.project_slug or 'clawker'}-{opts.agent}"[:63],
+            services=services,
             netns=effective_firewall,
             tty=opts.tty,
             autostart=opts.autostart,
@@ -187,16 +206,38 @@ class Orchestrator:
         return self.engine.start(name)
 
     def run(self, opts: RunOptions):
-        """create + start; returns SandboxInfo (agent CMD is already
-        spawning if autostart, else waiting for agent_ready)."""
+        """create + start (+ firewall gateway attach); returns SandboxInfo
+        (agent CMD is already spawning if autostart, else waiting for
+        agent_ready)."""
         info = self.create(opts)
         try:
-            return self.engine.start(info.name)
+            started = self.engine.start(info.name)
+            if info.labels.get("dev.clawker.fw") == "on":
+                self._fw_attach(info)
+            return started
         except BaseException:
             self.teardown(info.name, force=True)
             raise
 
+    def _fw_attach(self, info) -> None:
+        """Bind the host-side policy gateway for this sandbox via the CP
+        (reference: FirewallEnable enrolling the container post-start,
+        container_start.go:349)."""
+        from .controlplane.client import CPClient
+        cp = CPClient()
+        cp.request({"op": "fw_attach", "sandbox": info.name,
+                    "rundir": str(info.rundir)})
+
     def teardown(self, name: str, force: bool = False) -> None:
+        try:
+            from .controlplane.client import CPClient
+            cp = CPClient(auto_start=False)
+            if cp.running():
+                cp.request({"op": "fw_detach", "sandbox": name})
+        except ClawkerError:
+            pass
+        except Exception:
+            pass
         try:
             self.engine.remove(name, force=force)
         finally:

@@ -26,7 +26,12 @@ def isolated_env(tmp_path, monkeypatch):
         d = tmp_path / sub
         d.mkdir(exist_ok=True)
         monkeypatch.setenv(var, str(d))
-    return tmp_path
+    yield tmp_path
+    # the image store may have mounted a layer tmpfs (overlayfs nesting
+    # rule); unmount so pytest can clean its tmp dir
+    import subprocess
+    subprocess.run(["umount", "-l", str(tmp_path / "images" / "layers")],
+                   capture_output=True)
 
 
 def _isolation_available() -> bool:

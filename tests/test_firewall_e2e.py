@@ -1,0 +1,193 @@
+"""End-to-end egress firewall tests (ns backend): sandbox netns with no
+uplink + ckgw shims + host policy gateway via cpd.
+
+Reference analog: test/e2e/firewall_test.go:95-450 (blocked/allowed/
+path rules/bypass/add-remove against the real CP + Envoy + CoreDNS +
+eBPF stack)."""
+import http.server
+import json
+import socket
+import threading
+import time
+from pathlib import Path
+
+import pytest
+
+from conftest import requires_isolation
+
+pytestmark = requires_isolation
+
+
+class _Upstream(http.server.BaseHTTPRequestHandler):
+    def do_GET(self):
+        body = f"UPSTREAM_OK path={self.path}".encode()
+        self.send_response(200)
+        self.send_header("Content-Length", str(len(body)))
+        self.end_headers()
+        self.wfile.write(body)
+
+    def log_message(self, *a):
+        pass
+
+
+@pytest.fixture
+def upstream():
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), _Upstream)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    yield srv.server_address[1]
+    srv.shutdown()
+
+
+@pytest.fixture
+def fw_env(isolated_env, tmp_path, monkeypatch, upstream):
+    monkeypatch.setenv(
+        "CLAWKER_DNS_STATIC",
+        "allowed.test=127.0.0.1,denied.test=127.0.0.1,tunnel.test=127.0.0.1")
+    ws = tmp_path / "fwproj"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text("project: fwtest\n")
+    from clawker_amd.config import load_config
+    from clawker_amd.config.schema import EgressRule
+    from clawker_amd.firewall import EgressRulesStore
+    from clawker_amd.orchestrator import Orchestrator
+    store = EgressRulesStore()
+    store.add([
+        EgressRule(dst="allowed.test", proto="http", port=upstream,
+                   deny_paths=["/secret"]),
+        EgressRule(dst="tunnel.test", proto="tcp", port=upstream),
+    ])
+    orch = Orchestrator(load_config(ws))
+    yield orch, ws, upstream
+    for info in orch.engine.list():
+        try:
+            orch.teardown(info.name, force=True)
+        except Exception:
+            pass
+    from clawker_amd.controlplane.client import CPClient
+    CPClient(auto_start=False).stop()
+    orch.close()
+
+
+AGENT_SCRIPT = r"""
+import json, os, socket, urllib.request, urllib.error
+out = {}
+
+def get(url):
+    try:
+        with urllib.request.urlopen(url, timeout=10) as r:
+            return r.status, r.read().decode()
+    except urllib.error.HTTPError as e:
+        return e.code, e.read().decode()
+    except Exception as e:
+        return -1, str(e)
+
+out["allowed"] = get("http://allowed.test:%PORT%/ok")
+out["denied_domain"] = get("http://denied.test:%PORT%/")
+out["denied_path"] = get("http://allowed.test:%PORT%/secret/x")
+try:
+    socket.getaddrinfo("allowed.test", None)
+    out["dns_allowed"] = "ok"
+except OSError as e:
+    out["dns_allowed"] = f"fail:{e}"
+try:
+    socket.getaddrinfo("not-in-policy.test", None)
+    out["dns_denied"] = "resolved"   # BAD
+except OSError:
+    out["dns_denied"] = "nxdomain"
+# CONNECT tunnel through the proxy to an allowed tcp rule
+import http.client
+c = http.client.HTTPConnection("127.0.0.1", 3128, timeout=10)
+c.set_tunnel("tunnel.test", %PORT%)
+try:
+    c.request("GET", "/tunneled")
+    r = c.getresponse()
+    out["tunnel"] = (r.status, r.read().decode())
+except Exception as e:
+    out["tunnel"] = (-1, str(e))
+c2 = http.client.HTTPConnection("127.0.0.1", 3128, timeout=10)
+c2.set_tunnel("evil.test", %PORT%)
+try:
+    c2.request("GET", "/")
+    out["tunnel_denied"] = (c2.getresponse().status, "")
+except Exception as e:
+    out["tunnel_denied"] = (-1, str(e))
+print("RESULT " + json.dumps(out), flush=True)
+"""
+
+
+def _wait_gateway(orch, name, deadline=10.0):
+    """Wait until cpd has attached the gateway sockets for the sandbox."""
+    rundir = orch.engine.inspect(name).rundir
+    end = time.monotonic() + deadline
+    while time.monotonic() < end:
+        if (rundir / "egress.sock").exists() and (rundir / "dns.sock").exists():
+            return True
+        time.sleep(0.05)
+    return False
+
+
+def test_egress_policy_enforced_end_to_end(fw_env):
+    orch, ws, port = fw_env
+    from clawker_amd.orchestrator import RunOptions
+    script = AGENT_SCRIPT.replace("%PORT%", str(port))
+    name = "clawker.fwtest.agent"
+    info = orch.run(RunOptions(
+        agent="agent", name=name, autostart=False, firewall=True,
+        cmd=["python3", "-c", script]))
+    assert info.labels["dev.clawker.fw"] == "on"
+    assert _wait_gateway(orch, name)
+    with orch.client(name) as c:
+        c.agent_ready()
+    code = orch.engine.wait(name, timeout_s=60)
+    logs = orch.engine.logs(name).decode()
+    assert code == 0, logs
+    result = json.loads(logs.split("RESULT ", 1)[1].splitlines()[0])
+    # allowed domain + path
+    assert result["allowed"][0] == 200
+    assert "UPSTREAM_OK" in result["allowed"][1]
+    # unknown domain: denied at the proxy (403) or at DNS
+    assert result["denied_domain"][0] in (403, -1)
+    # deny_paths enforced on plain http
+    assert result["denied_path"][0] == 403
+    # DNS zone policy
+    assert result["dns_allowed"] == "ok"
+    assert result["dns_denied"] == "nxdomain"
+    # CONNECT tunnels: allowed rule works, unknown host refused
+    assert result["tunnel"][0] == 200
+    assert "/tunneled" in result["tunnel"][1]
+    assert result["tunnel_denied"][0] != 200
+
+    # decisions recorded in the CP event log
+    from clawker_amd.controlplane.client import CPClient
+    evs = CPClient().events(200)
+    kinds = {(e.get("action"), e.get("dst")) for e in evs
+             if e["event"] == "egress_decision"}
+    assert ("allow", "allowed.test") in kinds
+    assert ("deny", "denied.test") in kinds or ("nxdomain", "not-in-policy.test") in kinds
+    orch.teardown(name, force=True)
+
+
+def test_bypass_dead_man(fw_env):
+    orch, ws, port = fw_env
+    from clawker_amd.controlplane.client import CPClient
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.fwtest.bypass"
+    orch.run(RunOptions(agent="bypass", name=name, autostart=False, firewall=True,
+                        cmd=["python3", "-c", AGENT_SCRIPT.replace("%PORT%", str(port))]))
+    assert _wait_gateway(orch, name)
+    cp = CPClient()
+    cp.bypass(2)    # 2-second bypass
+    time.sleep(0.3)
+    rundir = orch.engine.inspect(name).rundir
+    pol = json.loads((rundir / "policy.json").read_text())
+    assert pol["bypass"] is True
+    # dead-man: restored after expiry
+    deadline = time.monotonic() + 10
+    while time.monotonic() < deadline:
+        pol = json.loads((rundir / "policy.json").read_text())
+        if pol["bypass"] is False:
+            break
+        time.sleep(0.2)
+    assert pol["bypass"] is False
+    orch.teardown(name, force=True)
