@@ -171,6 +171,12 @@ class Engine:
             dst = rundir / "bin" / binname
             if not dst.exists() or dst.stat().st_mtime < src.stat().st_mtime:
                 shutil.copy2(src, dst)
+        # host-service helper scripts (hostproxy clients)
+        assets = Path(__file__).resolve().parents[1] / "assets"
+        for helper in ("host-open.sh", "git-credential-clawker"):
+            src = assets / helper
+            if src.is_file():
+                shutil.copy2(src, rundir / "bin" / helper)
 
         # per-sandbox identity files
         (rundir / "hostname").write_text(spec.hostname + "\n")

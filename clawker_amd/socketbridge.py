@@ -1,0 +1,152 @@
+"""SSH/GPG agent forwarding into sandboxes.
+
+Reference: internal/socketbridge — a muxrpc protocol over `docker exec`
+proxying container Unix sockets to the host's $SSH_AUTH_SOCK / gpg-agent.
+Redesign: every sandbox already shares its rundir (/run/clawker inside),
+so the bridge just LISTENS in the rundir and relays to the host agent
+socket — no exec channel, no framing protocol, one thread per connection.
+"""
+from __future__ import annotations
+
+import os
+import socket
+import subprocess
+import threading
+from dataclasses import dataclass, field
+from pathlib import Path
+
+from .engine.wire import bind_unix, connect_unix
+from .logger import get as get_logger
+
+log = get_logger("socketbridge")
+
+SSH_SOCK_NAME = "ssh-agent.sock"
+GPG_SOCK_NAME = "gpg-agent.sock"
+
+
+def host_ssh_auth_sock() -> str | None:
+    p = os.environ.get("SSH_AUTH_SOCK")
+    return p if p and os.path.exists(p) else None
+
+
+def host_gpg_extra_sock() -> str | None:
+    try:
+        r = subprocess.run(["gpgconf", "--list-dirs", "agent-extra-socket"],
+                           capture_output=True, text=True, timeout=5)
+        p = r.stdout.strip()
+        return p if r.returncode == 0 and p and os.path.exists(p) else None
+    except (OSError, subprocess.TimeoutExpired):
+        return None
+
+
+@dataclass
+class _Bridge:
+    listener: socket.socket
+    target: str
+    stop: threading.Event = field(default_factory=threading.Event)
+
+
+class SocketBridgeManager:
+    """Per-sandbox agent-socket bridges (attach/detach like the firewall
+    gateway; driven by the CP daemon's watcher)."""
+
+    def __init__(self):
+        self._bridges: dict[str, list[_Bridge]] = {}
+        self._lock = threading.Lock()
+
+    def attach(self, name: str, rundir: Path) -> list[str]:
+        """Create bridges for every available host agent socket; returns
+        the in-sandbox socket names created."""
+        created: list[str] = []
+        pairs = []
+        ssh = host_ssh_auth_sock()
+        if ssh:
+            pairs.append((SSH_SOCK_NAME, ssh))
+        gpg = host_gpg_extra_sock()
+        if gpg:
+            pairs.append((GPG_SOCK_NAME, gpg))
+        with self._lock:
+            if name in self._bridges or not pairs:
+                return created
+            bridges = []
+            for sock_name, target in pairs:
+                path = rundir / sock_name
+                path.unlink(missing_ok=True)
+                try:
+                    lst = bind_unix(path)
+                except OSError as e:
+                    log.warn("bridge_bind_failed", sock=sock_name, err=str(e))
+                    continue
+                os.chmod(path, 0o666)
+                lst.listen(16)
+                lst.settimeout(0.5)
+                br = _Bridge(listener=lst, target=target)
+                threading.Thread(target=self._accept_loop, args=(br,),
+                                 daemon=True).start()
+                bridges.append(br)
+                created.append(sock_name)
+            self._bridges[name] = bridges
+        if created:
+            log.info("bridge_attached", sandbox=name, socks=created)
+        return created
+
+    def detach(self, name: str) -> None:
+        with self._lock:
+            bridges = self._bridges.pop(name, [])
+        for br in bridges:
+            br.stop.set()
+            try:
+                br.listener.close()
+            except OSError:
+                pass
+
+    def detach_all(self) -> None:
+        for name in list(self._bridges):
+            self.detach(name)
+
+    def _accept_loop(self, br: _Bridge) -> None:
+        while not br.stop.is_set():
+            try:
+                conn, _ = br.listener.accept()
+            except socket.timeout:
+                continue
+            except OSError:
+                return
+            threading.Thread(target=self._relay, args=(br, conn),
+                             daemon=True).start()
+
+    def _relay(self, br: _Bridge, conn: socket.socket) -> None:
+        try:
+            up = connect_unix(br.target, timeout=10)
+        except OSError as e:
+            log.warn("bridge_upstream_failed", err=str(e))
+            conn.close()
+            return
+        up.settimeout(None)
+        conn.settimeout(None)
+
+        def pump(src, dst):
+            try:
+                while True:
+                    data = src.recv(65536)
+                    if not data:
+                        break
+                    dst.sendall(data)
+            except OSError:
+                pass
+            finally:
+                for s in (src, dst):
+                    try:
+                        s.shutdown(socket.SHUT_RDWR)
+                    except OSError:
+                        pass
+
+        t = threading.Thread(target=pump, args=(up, conn), daemon=True)
+        t.start()
+        pump(conn, up)
+        t.join(timeout=5)
+        for s in (conn, up):
+            try:
+                s.close()
+            except OSError:
+                pass

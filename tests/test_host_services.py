@@ -1,0 +1,132 @@
+"""Hostproxy + socketbridge tests (reference: internal/hostproxy routes,
+internal/socketbridge forwarding)."""
+import json
+import os
+import socket
+import threading
+import time
+from pathlib import Path
+
+import pytest
+
+from conftest import requires_isolation
+
+
+@pytest.fixture
+def hostproxy(isolated_env):
+    from clawker_amd.hostproxy import HostProxyManager
+    mgr = HostProxyManager()
+    mgr.ensure_running()
+    yield mgr
+    mgr.stop()
+
+
+def test_hostproxy_healthz_and_idempotent_ensure(hostproxy):
+    status, body = hostproxy.request("GET", "/healthz")
+    assert status == 200
+    hostproxy.ensure_running()    # no second spawn
+    assert hostproxy.running()
+
+
+def test_open_url_policy_fail_closed(hostproxy, isolated_env):
+    # no rule for the domain -> 403
+    status, body = hostproxy.request(
+        "POST", "/open/url", json.dumps({"url": "https://evil.example/x"}).encode())
+    assert status == 403
+    # rule added -> allowed (no browser on CI: opened:false with hint)
+    from clawker_amd.config.schema import EgressRule
+    from clawker_amd.firewall import EgressRulesStore
+    EgressRulesStore().add([EgressRule(dst="docs.example", proto="tls", port=443,
+                                       deny_paths=["/private"])])
+    status, body = hostproxy.request(
+        "POST", "/open/url", json.dumps({"url": "https://docs.example/page"}).encode())
+    assert status == 200, body
+    # deny_path enforced on open/url too (mirrored semantics)
+    status, _ = hostproxy.request(
+        "POST", "/open/url",
+        json.dumps({"url": "https://docs.example/private/key"}).encode())
+    assert status == 403
+    # junk url
+    status, _ = hostproxy.request(
+        "POST", "/open/url", json.dumps({"url": "file:///etc/passwd"}).encode())
+    assert status == 400
+
+
+def test_callback_register_and_hit(hostproxy):
+    status, body = hostproxy.request(
+        "POST", "/callback/register",
+        json.dumps({"port": 8765, "sandbox": "clawker.t.a"}).encode())
+    assert status == 200
+    sess = json.loads(body)
+    sid = sess["session"]
+    assert f"/cb/{sid}/" in sess["callback_url"]
+    status, body = hostproxy.request("GET", f"/cb/{sid}/done?code=xyz")
+    assert status == 200 and b"authentication complete" in body
+    status, _ = hostproxy.request("GET", "/cb/bogus/done")
+    assert status == 404
+
+
+def test_socketbridge_relays_to_host_agent(isolated_env, tmp_path, monkeypatch):
+    """Fake host ssh-agent (unix echo) -> bridge in rundir -> client."""
+    host_agent = tmp_path / "agent.sock"
+    srv = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    srv.bind(str(host_agent))
+    srv.listen(4)
+
+    def serve():
+        while True:
+            try:
+                c, _ = srv.accept()
+            except OSError:
+                return
+            data = c.recv(1024)
+            c.sendall(b"AGENT:" + data)
+            c.close()
+
+    threading.Thread(target=serve, daemon=True).start()
+    monkeypatch.setenv("SSH_AUTH_SOCK", str(host_agent))
+
+    from clawker_amd.socketbridge import SocketBridgeManager
+    rundir = tmp_path / "rundir"
+    rundir.mkdir()
+    mgr = SocketBridgeManager()
+    created = mgr.attach("clawker.t.sb", rundir)
+    assert "ssh-agent.sock" in created
+    c = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    c.settimeout(5)
+    c.connect(str(rundir / "ssh-agent.sock"))
+    c.sendall(b"sign-request")
+    assert c.recv(100) == b"AGENT:sign-request"
+    c.close()
+    mgr.detach("clawker.t.sb")
+    srv.close()
+
+
+@requires_isolation
+def test_hostproxy_reachable_from_inside_sandbox(isolated_env, tmp_path, hostproxy):
+    """ns backend: the hostproxy socket is bind-mounted into the sandbox
+    and the host-open helper talks through it (policy-denied URL -> 403)."""
+    ws = tmp_path / "hsproj"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text("project: hstest\n")
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    orch = Orchestrator(load_config(ws))
+    name = "clawker.hstest.a"
+    try:
+        orch.run(RunOptions(
+            agent="a", name=name, autostart=True, firewall=True,
+            host_services=True,
+            cmd=["/bin/sh", "-c",
+                 "ls -la /run/clawker/hostproxy.sock; "
+                 "/run/clawker/bin/host-open.sh https://not-allowed.example/; "
+                 "echo open_rc=$?"]))
+        code = orch.engine.wait(name, timeout_s=30)
+        logs = orch.engine.logs(name).decode()
+        assert code == 0, logs
+        assert "hostproxy.sock" in logs
+        assert "open_rc=1" in logs       # denied by policy through the proxy
+        assert "not in egress policy" in logs
+    finally:
+        orch.teardown(name, force=True)
+        orch.close()
