@@ -177,6 +177,10 @@ class Engine:
             src = assets / helper
             if src.is_file():
                 shutil.copy2(src, rundir / "bin" / helper)
+        # per-agent bootstrap identity material (reference:
+        # InstallAgentBootstrapMaterial, agent_bootstrap.go)
+        from ..auth import install_bootstrap
+        install_bootstrap(rundir, spec.name)
 
         # per-sandbox identity files
         (rundir / "hostname").write_text(spec.hostname + "\n")

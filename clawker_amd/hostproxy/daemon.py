@@ -163,7 +163,14 @@ class HostProxyHandlerBase(BaseHTTPRequestHandler):
     def _git_credential(self) -> None:
         """`git credential fill` against the HOST credential store —
         credentials are never copied into sandboxes (reference:
-        containerfs.go:1-12 doctrine + hostproxy git route)."""
+        containerfs.go:1-12 doctrine + hostproxy git route). Requires a
+        valid agent bootstrap token (the per-agent cert analog)."""
+        from ..auth import verify_agent_token
+        token = self.headers.get("X-Clawker-Token", "").strip()
+        if not token or verify_agent_token(token) is None:
+            log.warn("git_credential_denied", reason="bad agent token")
+            self._json(403, {"error": "invalid agent token"})
+            return
         try:
             payload = self._body().decode()
             r = subprocess.run(["git", "credential", "fill"], input=payload,
