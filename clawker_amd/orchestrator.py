@@ -120,6 +120,24 @@ class Orchestrator:
             else:
                 mounts.append(Mount(src=str(ws_src), dst=ws_dst))
                 ws_effective = ws_src if backend == "proc" else Path(ws_dst)
+                # worktree workspaces: the .git FILE points at the parent
+                # repo's gitdir, which the sandbox can't see (host /tmp etc.
+                # are masked) — read-through mount the main .git at its host
+                # path (reference: buildWorktreeGitMounts, setup.go:306)
+                gitfile = Path(ws_src) / ".git"
+                if backend == "ns" and gitfile.is_file():
+                    try:
+                        ref = gitfile.read_text().strip()
+                        if ref.startswith("gitdir:"):
+                            gitdir = Path(ref.split(":", 1)[1].strip())
+                            common = gitdir
+                            # .git/worktrees/<name> -> the main .git dir
+                            if common.parent.name == "worktrees":
+                                common = common.parent.parent
+                            if common.is_dir():
+                                mounts.append(Mount(src=str(common), dst=str(common)))
+                    except OSError:
+                        pass
         if proj.workspace.share_volume and self.cfg.project_slug:
             share, _ = self.engine.ensure_volume(
                 f"clawker.{self.cfg.project_slug}.share", {consts.MANAGED_LABEL: "true"})

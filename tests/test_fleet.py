@@ -1,0 +1,77 @@
+"""Fleet fan-out tests: N parallel agent loops over worktrees (CPU; the
+GPU pinning branch is covered by allocator tests + gpu-marked tests)."""
+import json
+import subprocess
+import time
+
+import pytest
+
+from conftest import requires_isolation
+
+pytestmark = requires_isolation
+
+
+@pytest.fixture
+def git_proj(isolated_env, tmp_path):
+    root = tmp_path / "fleetproj"
+    root.mkdir()
+    (root / ".clawker.yaml").write_text(
+        "project: fleettest\nagent:\n  harness: echo\n")
+    (root / "work.txt").write_text("base\n")
+    subprocess.run(["git", "init", "-q", "-b", "main"], cwd=root, check=True)
+    subprocess.run(["git", "add", "-A"], cwd=root, check=True)
+    subprocess.run(["git", "-c", "user.email=t@t", "-c", "user.name=t",
+                    "commit", "-q", "-m", "init"], cwd=root, check=True)
+    return root
+
+
+def test_fleet_up_wait_down_with_worktrees(git_proj):
+    from clawker_amd.config import load_config
+    from clawker_amd.fleet import Fleet, FleetOptions
+    cfg = load_config(git_proj)
+    fleet = Fleet(cfg)
+    members = fleet.up(FleetOptions(
+        count=3, firewall=False,
+        cmd=["/bin/sh", "-c",
+             "echo agent=$CLAWKER_AGENT branch=$(git branch --show-current); "
+             "echo done-$CLAWKER_AGENT >> out.txt"]))
+    assert len(members) == 3
+    assert {m.branch for m in members} == {"agent/0", "agent/1", "agent/2"}
+    fleet.wait(members, timeout_s=60)
+    assert all(m.exit_code == 0 for m in members), [
+        (m.sandbox, m.exit_code, fleet.orch.engine.logs(m.sandbox)[-200:])
+        for m in members]
+    # each agent worked in ITS OWN worktree
+    for i, m in enumerate(members):
+        logs = fleet.orch.engine.logs(m.sandbox).decode()
+        assert f"branch=agent/{i}" in logs
+    # the base repo is untouched
+    assert (git_proj / "work.txt").read_text() == "base\n"
+    assert not (git_proj / "out.txt").exists()
+    # worktree dirs got the agent outputs
+    from clawker_amd.project.worktrees import worktrees_dir
+    for i in range(3):
+        wt = worktrees_dir(cfg) / f"agent-{i}"
+        assert (wt / "out.txt").read_text() == f"done-agent{i}\n"
+    n = fleet.down()
+    assert n == 3
+    assert fleet.status() == []
+    fleet.orch.close()
+
+
+def test_fleet_snapshot_mode_without_git(isolated_env, tmp_path):
+    root = tmp_path / "nogit"
+    root.mkdir()
+    (root / ".clawker.yaml").write_text("project: nogit\nagent:\n  harness: echo\n")
+    (root / "seed.txt").write_text("s")
+    from clawker_amd.config import load_config
+    from clawker_amd.fleet import Fleet, FleetOptions
+    fleet = Fleet(load_config(root))
+    members = fleet.up(FleetOptions(
+        count=2, firewall=False,
+        cmd=["/bin/sh", "-c", "echo x >> seed.txt; cat seed.txt"]))
+    fleet.wait(members, timeout_s=60)
+    assert all(m.exit_code == 0 for m in members)
+    assert (root / "seed.txt").read_text() == "s"   # snapshots, not binds
+    fleet.down()
+    fleet.orch.close()
