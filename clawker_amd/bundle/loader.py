@@ -42,6 +42,7 @@ class Harness:
     managed_prompt_path: str = ""
     egress: list[EgressRule] = field(default_factory=list, metadata={"merge": "union"})
     seeds: dict = field(default_factory=dict)      # in-image path -> content
+    staging: list[dict] = field(default_factory=list)   # host-state staging
     post_init: str = ""     # script body (InitPlan)
     pre_run: str = ""       # script body (BootPlan)
 

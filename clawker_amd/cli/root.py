@@ -44,7 +44,7 @@ def cli():
 
 
 def main() -> int:
-    from . import container, firewall, fleet, image, monitor, project, settings, cp, worktree  # noqa
+    from . import container, firewall, fleet, image, monitor, project, settings, cp, volume, worktree  # noqa
     try:
         cli(standalone_mode=False)
         return 0
@@ -68,7 +68,7 @@ def main() -> int:
 
 
 # import groups at module load so `clawker --help` lists them
-from . import container, firewall, fleet, image, monitor, project, settings, cp, worktree  # noqa: E402,F401
+from . import container, firewall, fleet, image, monitor, project, settings, cp, volume, worktree  # noqa: E402,F401
 
 if __name__ == "__main__":
     # re-enter through the canonical module path so command registration

@@ -1,0 +1,79 @@
+"""Volume + auth verbs (reference: internal/cmd/volume, `clawker auth
+rotate`)."""
+from __future__ import annotations
+
+import json
+import time
+
+import click
+
+from ..cmdutil import format_age
+from ..errors import ClawkerError
+from .root import Ctx, cli, pass_factory
+
+
+@cli.group("volume")
+def volume_group():
+    """Named volumes (agent config/history state; survive sandbox rm)."""
+
+
+@volume_group.command("ls")
+@click.option("--format", "fmt", default="")
+@pass_factory
+def volume_ls(ctx: Ctx, fmt):
+    f = ctx.factory
+    vols = f.engine().db.list_volumes()
+    if fmt == "json":
+        f.io.print(json.dumps(vols, indent=1))
+        return
+    from rich.table import Table
+    t = Table(box=None, pad_edge=False)
+    for c in ("NAME", "PATH", "AGE"):
+        t.add_column(c)
+    now = time.time()
+    for v in vols:
+        t.add_row(v["name"], v["path"], format_age(now - v["created"]))
+    f.io.print(t)
+
+
+@volume_group.command("rm")
+@click.argument("names", nargs=-1, required=True)
+@pass_factory
+def volume_rm(ctx: Ctx, names):
+    f = ctx.factory
+    for n in names:
+        f.engine().remove_volume(n)
+        f.io.print(n)
+
+
+@volume_group.command("prune")
+@pass_factory
+def volume_prune(ctx: Ctx):
+    """Remove volumes not referenced by any sandbox."""
+    f = ctx.factory
+    eng = f.engine()
+    live = {i.name for i in eng.list()}
+    removed = 0
+    for v in eng.db.list_volumes():
+        owner = v["name"].rsplit("-", 1)[0]
+        if v["name"].startswith("clawker.") and owner not in live \
+                and not v["name"].endswith(".share"):
+            eng.remove_volume(v["name"])
+            f.io.print(v["name"])
+            removed += 1
+    f.io.eprint(f"removed {removed} volume(s)")
+
+
+@cli.group("auth")
+def auth_group():
+    """Agent identity trust material."""
+
+
+@auth_group.command("rotate")
+@pass_factory
+def auth_rotate(ctx: Ctx):
+    """Rotate the root key; existing agent tokens become invalid
+    (restart sandboxes to re-mint). Reference: `clawker auth rotate`."""
+    from .. import auth
+    auth.rotate_auth_material()
+    ctx.factory.io.success("auth material rotated (restart agents to re-mint tokens)")

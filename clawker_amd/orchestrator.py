@@ -144,6 +144,35 @@ class Orchestrator:
             mounts.append(Mount(src=str(share), dst="/share"))
         mounts.extend(opts.mounts)
 
+        # harness config volumes + host-state staging (reference:
+        # EnsureConfigVolumes + containerfs staging manifest)
+        harness = None
+        try:
+            from .bundle import load_harness
+            harness = load_harness(proj.agent.harness, self.cfg.project_root)
+        except ClawkerError:
+            pass
+        if harness is not None and backend == "ns":
+            h_user = harness.user or "root"
+            home = "/root" if h_user in ("", "root") else f"/home/{h_user}"
+            for vol in harness.config_volumes:
+                vol_name = f"{name}-{vol.lstrip('.')}"
+                vol_path, fresh = self.engine.ensure_volume(
+                    vol_name, {consts.MANAGED_LABEL: "true"})
+                if fresh and harness.staging:
+                    from .containerfs import stage_host_state
+                    relevant = [e for e in harness.staging
+                                if str(e.get("dst", "")).startswith(vol + "/")
+                                or str(e.get("dst", "")) == vol]
+                    # entries are HOME-relative; this volume holds <vol>/
+                    staged = stage_host_state(
+                        [{**e, "dst": str(e["dst"])[len(vol):].lstrip("/") or "."}
+                         for e in relevant], vol_path)
+                    if staged:
+                        log.info("host_state_staged", sandbox=name,
+                                 volume=vol, files=len(staged))
+                mounts.append(Mount(src=str(vol_path), dst=f"{home}/{vol}"))
+
         # -- env contract (SURVEY.md A.1 + GPU vars) -------------------------
         env: dict[str, str] = {
             consts.ENV_PROJECT: self.cfg.project_slug,
@@ -164,6 +193,11 @@ class Orchestrator:
                 env["CLAWKER_HBM_GB"] = str(hbm)
                 pct = max(1, min(100, round(hbm * 100 / 288)))
                 env["GPU_MAX_ALLOC_PERCENT"] = str(pct)
+        if proj.agent.env_file:
+            from .dotenv import parse_env_file
+            base_dir = self.cfg.project_root or Path.cwd()
+            env.update(parse_env_file(
+                (base_dir / proj.agent.env_file).expanduser()))
         env.update(proj.agent.env)
         env.update(opts.env)
 
