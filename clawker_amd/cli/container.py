@@ -481,3 +481,55 @@ def prune_cmd(ctx: Ctx, force):
             f.io.print(i.name)
             removed += 1
     f.io.eprint(f"removed {removed} sandbox(es)")
+
+
+@cli.command("cp")
+@click.argument("src")
+@click.argument("dst")
+@pass_factory
+def cp_cmd(ctx: Ctx, src, dst):
+    """Copy files between the host and a running sandbox
+    (SANDBOX:PATH <-> PATH), streamed as tar through the control socket."""
+    import io
+    import tarfile
+    from pathlib import Path as _P
+    f = ctx.factory
+
+    def split(ref):
+        if ":" in ref and not ref.startswith("/") and not ref.startswith("."):
+            name, _, path = ref.partition(":")
+            return resolve_sandbox_name(f, name), path
+        return None, ref
+
+    src_sb, src_path = split(src)
+    dst_sb, dst_path = split(dst)
+    if (src_sb is None) == (dst_sb is None):
+        raise ClawkerError("exactly one of SRC/DST must be SANDBOX:PATH")
+    if src_sb:
+        parent = str(_P(src_path).parent) or "/"
+        base = _P(src_path).name
+        code, out, err = f.engine().exec(
+            src_sb, ["tar", "-C", parent, "-cf", "-", base])
+        if code != 0:
+            raise ClawkerError(f"tar in sandbox failed: {err.decode()[-200:]}")
+        with tarfile.open(fileobj=io.BytesIO(out)) as tf:
+            dest = _P(dst_path)
+            dest.mkdir(parents=True, exist_ok=True)
+            tf.extractall(dest)
+        f.io.success(f"copied {src} -> {dst}")
+    else:
+        sp = _P(src_path)
+        if not sp.exists():
+            raise ClawkerError(f"no such file: {sp}")
+        buf = io.BytesIO()
+        with tarfile.open(fileobj=buf, mode="w") as tf:
+            tf.add(str(sp), arcname=sp.name)
+        code, out, err = f.engine().exec(
+            dst_sb, ["/bin/sh", "-c", f"mkdir -p {dst_path} && tar -C {dst_path} -xf -"],
+            stdin=buf.getvalue())
+        if code != 0:
+            raise ClawkerError(f"tar extract failed: {err.decode()[-200:]}")
+        f.io.success(f"copied {src} -> {dst}")
+
+
+container_group.add_command(cp_cmd, "cp")

@@ -9,7 +9,7 @@ import click
 from .root import Ctx, cli, pass_factory
 
 
-@cli.group("cp")
+@cli.group("controlplane")
 def cp_group():
     """Control-plane daemon (cpd)."""
 

@@ -83,3 +83,60 @@ def image_prune(ctx: Ctx):
     f = ctx.factory
     n = f.engine().images.prune_layers()
     f.io.eprint(f"removed {n} unreferenced layer(s)")
+
+
+@cli.group("harness")
+def harness_group():
+    """Agent harness bundles (claude, codex, ...)."""
+
+
+@harness_group.command("list")
+@pass_factory
+def harness_list(ctx: Ctx):
+    from ..bundle import list_harnesses, load_harness
+    f = ctx.factory
+    cfg = f.config()
+    from rich.table import Table
+    t = Table(box=None, pad_edge=False)
+    for c in ("NAME", "CMD", "STACKS", "EGRESS RULES"):
+        t.add_column(c)
+    for name in list_harnesses(cfg.project_root):
+        try:
+            h = load_harness(name, cfg.project_root)
+            t.add_row(name, " ".join(h.cmd)[:40], ",".join(h.stacks) or "-",
+                      str(len(h.egress)))
+        except Exception:
+            t.add_row(name, "?", "?", "?")
+    f.io.print(t)
+
+
+@harness_group.command("show")
+@click.argument("name")
+@pass_factory
+def harness_show(ctx: Ctx, name):
+    from ..bundle import load_harness
+    from ..storage.store import to_plain
+    h = load_harness(name, ctx.factory.config().project_root)
+    ctx.factory.io.print(json.dumps(to_plain(h), indent=1))
+
+
+@cli.group("stack")
+def stack_group():
+    """Language/toolchain stacks baked into base images."""
+
+
+@stack_group.command("list")
+@pass_factory
+def stack_list(ctx: Ctx):
+    from pathlib import Path
+    from ..bundle.loader import ASSETS, load_stack
+    f = ctx.factory
+    names = sorted(p.stem for p in (ASSETS / "stacks").glob("*.yaml"))
+    from rich.table import Table
+    t = Table(box=None, pad_edge=False)
+    for c in ("NAME", "PACKAGES", "DESCRIPTION"):
+        t.add_column(c)
+    for n in names:
+        s = load_stack(n)
+        t.add_row(n, ",".join(s.packages) or "-", s.description)
+    f.io.print(t)

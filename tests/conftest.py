@@ -11,6 +11,18 @@ def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires a real MI355X GPU (run via gpurun)")
 
 
+@pytest.fixture(scope="session", autouse=True)
+def _ensure_native_built():
+    """Most tests drive real sandboxes; build the native runtime once."""
+    import subprocess
+    root = Path(__file__).resolve().parent.parent
+    missing = [b for b in ("ckrt", "ckd", "ckgw")
+               if not (root / "native" / "bin" / b).exists()]
+    if missing or not list(root.glob("clawker_amd/_native*.so")):
+        subprocess.run(["make", "native", "pymod", "-j4"], cwd=root,
+                       check=True, capture_output=True)
+
+
 @pytest.fixture
 def isolated_env(tmp_path, monkeypatch):
     """Fully isolated clawker state dirs (reference: internal/testenv)."""

@@ -17,11 +17,7 @@ from clawker_amd.errors import ConflictError, NotFoundError
 from conftest import requires_isolation
 
 
-@pytest.fixture(autouse=True)
-def _build_native():
-    root = Path(__file__).resolve().parent.parent
-    if not (root / "native/bin/ckrt").exists() or not (root / "native/bin/ckd").exists():
-        subprocess.run(["make", "native"], cwd=root, check=True, capture_output=True)
+# native binaries are built once per session by the conftest fixture
 
 
 @pytest.fixture
