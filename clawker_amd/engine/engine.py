@@ -131,6 +131,36 @@ class Engine:
         self.store_root.mkdir(parents=True, exist_ok=True)
         self._passthrough: list[Mount] | None = None
         self.backend = detect_backend()
+        if self.backend == "ns":
+            self._ensure_upper_filesystem()
+
+    def _ensure_upper_filesystem(self) -> None:
+        """Overlay upperdirs cannot live on an overlayfs. Hosts whose root
+        is itself an overlay (CI containers) get a tmpfs mounted over the
+        sandbox store; a production node with a real filesystem is
+        untouched."""
+        try:
+            target = self.store_root.resolve()
+            best_len = -1
+            fstype = ""
+            for line in Path("/proc/self/mounts").read_text().splitlines():
+                parts = line.split()
+                if len(parts) < 3:
+                    continue
+                mp = parts[1].encode().decode("unicode_escape")
+                if (str(target) == mp or str(target).startswith(mp.rstrip("/") + "/")) \
+                        and len(mp) > best_len:
+                    best_len = len(mp)
+                    fstype = parts[2]
+            if fstype != "overlay":
+                return
+            r = subprocess.run(
+                ["mount", "-t", "tmpfs", "-o", "mode=700", "clawker-sandboxes",
+                 str(self.store_root)], capture_output=True, text=True)
+            if r.returncode != 0:
+                log.warn("upperfs_unavailable", err=r.stderr.strip())
+        except OSError:
+            pass
 
     def close(self) -> None:
         self.db.close()
