@@ -1,0 +1,147 @@
+"""Restart, concurrency and failure-path robustness (reference: CP
+resilience contract + ReapFailedStart + reconcile behaviors)."""
+import json
+import subprocess
+import sys
+import threading
+import time
+from pathlib import Path
+
+import pytest
+
+from conftest import requires_isolation
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+@pytest.fixture
+def orch(isolated_env, tmp_path):
+    ws = tmp_path / "rproj"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text("project: rtest\n")
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator
+    o = Orchestrator(load_config(ws))
+    yield o
+    for info in o.engine.list():
+        try:
+            o.teardown(info.name, force=True)
+        except Exception:
+            pass
+    o.close()
+
+
+@requires_isolation
+def test_sandbox_restart_cycle(orch):
+    """stop -> start -> stop again: run state cleared, init marker
+    persists (one-time InitPlan semantics across restarts)."""
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.rtest.re"
+    opts = RunOptions(agent="re", name=name, cmd=["/bin/sh", "-c", "echo run-$$; sleep 30"])
+    opts.autostart = False
+    orch.run(opts)
+    with orch.client(name) as c:
+        c.agent_initialized()
+        c.agent_ready()
+    orch.engine.stop(name)
+    assert orch.engine.inspect(name).state == "exited"
+    # restart the SAME sandbox
+    orch.engine.start(name)
+    with orch.client(name) as c:
+        h = c.hello()
+        assert h["initialized"] is True     # marker survived the restart
+        assert h["cmd_running"] is False
+        c.agent_ready()
+    info = orch.engine.inspect(name)
+    assert info.state == "running"
+    code = orch.engine.stop(name)
+    assert code is not None
+
+
+@requires_isolation
+def test_concurrent_exec_clients(orch):
+    """Multiple control connections exec simultaneously (ckd poll loop
+    multiplexing)."""
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.rtest.cc"
+    orch.run(RunOptions(agent="cc", name=name, autostart=True,
+                        cmd=["sleep", "30"]))
+    results = []
+    errors = []
+
+    def worker(i):
+        try:
+            code, out, _ = orch.engine.exec(
+                name, ["/bin/sh", "-c", f"echo w{i}; sleep 0.{i % 3}; echo d{i}"])
+            results.append((i, code, out))
+        except Exception as e:
+            errors.append((i, e))
+
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=30)
+    assert not errors, errors
+    assert len(results) == 6
+    for i, code, out in results:
+        assert code == 0
+        assert f"w{i}".encode() in out and f"d{i}".encode() in out
+    orch.engine.stop(name)
+
+
+@requires_isolation
+def test_failed_start_reclaims_gpu_allocations(orch, monkeypatch):
+    """create-scope reclaim: a failing start releases the sandbox's GPUs
+    (reference: createScope + ReapFailedStart)."""
+    from clawker_amd.gpu.inventory import GPUDevice, GPUInventory
+    from clawker_amd.orchestrator import RunOptions
+    # fake a GPU so allocation happens on this CPU host
+    fake = GPUInventory([GPUDevice(index=0, render_minor=128, card=0)])
+    orch.allocator.inventory = fake
+    # sabotage the runtime so start fails
+    monkeypatch.setenv("CLAWKER_NATIVE_BIN", "/nonexistent")
+    with pytest.raises(Exception):
+        orch.run(RunOptions(agent="fail", name="clawker.rtest.fail",
+                            gpus=1, cmd=["true"]))
+    assert orch.allocator.allocations() == {}
+    assert orch.engine.db.get_sandbox("clawker.rtest.fail") is None
+
+
+@requires_isolation
+def test_cp_survives_subsystem_errors(orch, isolated_env):
+    """cpd keeps serving after an op raises (no-exit-after-ready)."""
+    from clawker_amd.controlplane.client import CPClient
+    cp = CPClient()
+    cp.ensure_running()
+    try:
+        # bad op payload -> error response, daemon stays up
+        with pytest.raises(Exception):
+            cp.request({"op": "fw_attach"})   # missing fields
+        assert cp.status()["ready"] is True
+        with pytest.raises(Exception):
+            cp.request({"op": "nonsense"})
+        assert cp.status()["ready"] is True
+    finally:
+        cp.stop()
+
+
+@requires_isolation
+def test_stale_state_cleared_on_restart_after_kill(orch):
+    """SIGKILL the shim+sandbox, then start again cleanly."""
+    import os
+    import signal as sig
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.rtest.k9"
+    orch.run(RunOptions(agent="k9", name=name, autostart=True, cmd=["sleep", "60"]))
+    info = orch.engine.inspect(name)
+    os.kill(info.pid, sig.SIGKILL)
+    deadline = time.monotonic() + 10
+    while time.monotonic() < deadline:
+        if orch.engine.inspect(name).state != "running":
+            break
+        time.sleep(0.05)
+    assert orch.engine.inspect(name).state in ("exited", "dead")
+    orch.engine.start(name)
+    assert orch.engine.inspect(name).state == "running"
+    orch.engine.stop(name)
