@@ -43,10 +43,44 @@ def cli():
     logger.setup(consts.log_dir() / "clawker.log")
 
 
+def expand_user_alias(argv: list[str]) -> list[str]:
+    """User-defined alias expansion with $1..$N / $@ substitution
+    (reference: internal/cmd/root/useraliases.go). Only the first token
+    is eligible and built-ins always win."""
+    if not argv or argv[0].startswith("-") or argv[0] in cli.commands:
+        return argv
+    try:
+        from ..config.config import load_settings
+        aliases = load_settings().get().aliases
+    except Exception:
+        return argv
+    expansion = aliases.get(argv[0])
+    if not expansion:
+        return argv
+    rest = argv[1:]
+    out: list[str] = []
+    used = set()
+    import shlex
+    for tok in shlex.split(str(expansion)):
+        if tok == "$@":
+            out.extend(rest)
+            used.update(range(len(rest)))
+        elif len(tok) >= 2 and tok[0] == "$" and tok[1:].isdigit():
+            idx = int(tok[1:]) - 1
+            if idx >= len(rest):
+                raise FlagError(f"alias '{argv[0]}' needs argument ${tok[1:]}")
+            out.append(rest[idx])
+            used.add(idx)
+        else:
+            out.append(tok)
+    out.extend(a for i, a in enumerate(rest) if i not in used and "$" not in expansion)
+    return out
+
+
 def main() -> int:
     from . import container, firewall, fleet, image, monitor, project, settings, cp, volume, worktree  # noqa
     try:
-        cli(standalone_mode=False)
+        cli(args=expand_user_alias(sys.argv[1:]), standalone_mode=False)
         return 0
     except click.exceptions.Abort:
         return 130
