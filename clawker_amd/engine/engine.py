@@ -211,6 +211,11 @@ class Engine:
         # InstallAgentBootstrapMaterial, agent_bootstrap.go)
         from ..auth import install_bootstrap
         install_bootstrap(rundir, spec.name)
+        # MITM trust bundle staged into the rundir when the orchestrator
+        # minted one (label carries the host path)
+        bundle = spec.labels.pop("dev.clawker.trustbundle", None)
+        if bundle and Path(bundle).is_file():
+            shutil.copy2(bundle, rundir / "trust-bundle.crt")
 
 
         # rootfs stack from the image + host passthrough binds
