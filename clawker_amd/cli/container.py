@@ -533,3 +533,20 @@ def cp_cmd(ctx: Ctx, src, dst):
 
 
 container_group.add_command(cp_cmd, "cp")
+
+
+@cli.command("restart")
+@click.option("-t", "--time", "timeout", type=float, default=10.0, show_default=True)
+@click.argument("names", nargs=-1, required=True)
+@pass_factory
+def restart_cmd(ctx: Ctx, timeout, names):
+    """Stop then start sandboxes (agent CMD released immediately)."""
+    f = ctx.factory
+    for n in names:
+        name = resolve_sandbox_name(f, n)
+        f.engine().stop(name, timeout_s=timeout)
+        _boot_and_wait(f, name, interactive=False, tty=False, detach=True, rm=False)
+        f.io.print(name)
+
+
+container_group.add_command(restart_cmd, "restart")
