@@ -74,16 +74,21 @@ def boot_plan() -> list[Step]:
     ]
 
 
-def run_boot_plans(factory, name: str, client: CkdClient, hello: dict,
-                   quiet: bool = False) -> None:
+def drive_boot(client: CkdClient, hello: dict,
+               on_step: Callable[[str, int], None] | None = None) -> None:
     """Drive InitPlan (first boot only) then BootPlan, matching the
     reference's CP dial flow (SURVEY.md §3.1 lower half)."""
+    if not hello.get("initialized"):
+        run_plan(client, init_plan(), on_step)
+        client.agent_initialized()
+    run_plan(client, boot_plan(), on_step)
+
+
+def run_boot_plans(factory, name: str, client: CkdClient, hello: dict,
+                   quiet: bool = False) -> None:
     def on_step(step: str, code: int) -> None:
         if not quiet:
             mark = "✓" if code == 0 else "!"
             factory.io.eprint(f"  {mark} {step}")
 
-    if not hello.get("initialized"):
-        run_plan(client, init_plan(), on_step)
-        client.agent_initialized()
-    run_plan(client, boot_plan(), on_step)
+    drive_boot(client, hello, None if quiet else on_step)

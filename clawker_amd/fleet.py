@@ -62,7 +62,7 @@ class Fleet:
             ropts = RunOptions(
                 agent=agent, cmd=list(opts.cmd), env=dict(opts.env),
                 image=opts.image, gpus=opts.gpus_per_agent,
-                firewall=opts.firewall, autostart=True)
+                firewall=opts.firewall, autostart=False)
             branch = ""
             if use_wt:
                 wt = setup_worktree(self.cfg, f"{opts.branch_prefix}/{i}", opts.base)
@@ -72,6 +72,12 @@ class Fleet:
                 # disposable copies so parallel agents never collide
                 ropts.workspace_mode = "snapshot"
             info = self.orch.run(ropts)
+            # drive init/boot plans then release the CMD (reference: the CP
+            # Executor path — fan-out members get the same boot contract)
+            from .controlplane.plans import drive_boot
+            with self.orch.client(info.name) as c:
+                drive_boot(c, c.hello())
+                c.agent_ready()
             members.append(FleetMember(
                 agent=agent, sandbox=info.name, branch=branch, gpus=info.gpus,
                 state=info.state))

@@ -139,16 +139,66 @@ std::vector<char*> to_argv(const std::vector<std::string>& v) {
 
 struct Client {
   int fd;
-  std::string buf;          // partial frame bytes
+  std::string buf;          // partial inbound frame bytes
+  std::string out;          // pending outbound bytes (nonblocking writes)
   bool attached = false;    // subscribed to console stream
+  bool dead = false;
 };
 
 std::vector<Client> g_clients;
 
+constexpr size_t kMaxClientBacklog = 8 * 1024 * 1024;
+
+// Non-blocking framed send with per-client backlog: a stalled control
+// client must never block PID 1's poll loop (reference: pubsub bounded
+// buffers + drop; here the whole frame is buffered or the client is
+// marked dead when its backlog cap is exceeded).
+void send_to_client(Client& c, const mj::Value& v) {
+  if (c.dead) return;
+  std::string body = v.dump();
+  uint8_t hdr[4] = {
+      static_cast<uint8_t>(body.size() >> 24), static_cast<uint8_t>(body.size() >> 16),
+      static_cast<uint8_t>(body.size() >> 8), static_cast<uint8_t>(body.size())};
+  std::string frame(reinterpret_cast<char*>(hdr), 4);
+  frame += body;
+  if (c.out.empty()) {
+    ssize_t n = send(c.fd, frame.data(), frame.size(), MSG_DONTWAIT | MSG_NOSIGNAL);
+    if (n == static_cast<ssize_t>(frame.size())) return;
+    if (n < 0) {
+      if (errno != EAGAIN && errno != EWOULDBLOCK) {
+        c.dead = true;
+        return;
+      }
+      n = 0;
+    }
+    c.out.append(frame.data() + n, frame.size() - n);
+  } else {
+    c.out += frame;
+  }
+  if (c.out.size() > kMaxClientBacklog) c.dead = true;
+}
+
+void flush_client(Client& c) {
+  if (c.dead || c.out.empty()) return;
+  ssize_t n = send(c.fd, c.out.data(), c.out.size(), MSG_DONTWAIT | MSG_NOSIGNAL);
+  if (n > 0) c.out.erase(0, n);
+  else if (n < 0 && errno != EAGAIN && errno != EWOULDBLOCK) c.dead = true;
+}
+
+Client* find_client(int fd) {
+  for (auto& c : g_clients)
+    if (c.fd == fd && !c.dead) return &c;
+  return nullptr;
+}
+
+void send_to_fd(int fd, const mj::Value& v) {
+  if (Client* c = find_client(fd)) send_to_client(*c, v);
+}
+
 void broadcast(const mj::Value& v, bool attached_only = false) {
   for (auto& c : g_clients) {
     if (attached_only && !c.attached) continue;
-    ck::send_frame(c.fd, v);
+    send_to_client(c, v);
   }
 }
 
@@ -300,7 +350,7 @@ void start_exec(Client& cl, const mj::Value& req) {
   const auto& stages = req["stages"].as_arr();
   if (stages.empty()) {
     mj::Value e; e.set("t", "error").set("id", job.id).set("msg", "no stages");
-    ck::send_frame(cl.fd, e);
+    send_to_client(cl, e);
     return;
   }
   std::string init_stdin;
@@ -379,7 +429,7 @@ void start_exec(Client& cl, const mj::Value& req) {
 
   mj::Value started;
   started.set("t", "started").set("id", job.id);
-  ck::send_frame(cl.fd, started);
+  send_to_client(cl, started);
   g_execs[job.id] = job;
 }
 
@@ -394,7 +444,7 @@ bool pump_exec_fd(ExecJob& job, int which) {
       mj::Value out;
       out.set("t", "out").set("id", job.id).set("stream", (int64_t)which)
          .set("data", ck::b64_encode(reinterpret_cast<uint8_t*>(buf), n));
-      ck::send_frame(job.client_fd, out);
+      send_to_fd(job.client_fd, out);
       continue;
     }
     if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return true;
@@ -412,7 +462,7 @@ void maybe_finish_exec(ExecJob& job) {
     if (c < 0) return;
   mj::Value done;
   done.set("t", "done").set("id", job.id).set("code", (int64_t)job.codes.back());
-  ck::send_frame(job.client_fd, done);
+  send_to_fd(job.client_fd, done);
   g_execs.erase(job.id);
 }
 
@@ -452,7 +502,7 @@ void reap() {
           mj::Value se;
           se.set("t", "stage_exit").set("id", job.id)
             .set("idx", (int64_t)i).set("code", (int64_t)code);
-          ck::send_frame(job.client_fd, se);
+          send_to_fd(job.client_fd, se);
         }
       }
     }
@@ -494,25 +544,25 @@ void handle_frame(Client& cl, const mj::Value& req) {
      .set("cmd_running", g_agent.running)
      .set("pid", (int64_t)(g_agent.running ? g_agent.pid : -1))
      .set("version", "0.1.0");
-    ck::send_frame(cl.fd, r);
+    send_to_client(cl, r);
   } else if (t == "agent_ready") {
     // CP releases the user CMD (reference: boot_steps.go AgentReady)
     spawn_agent(req["cmd"]);
     mj::Value r;
     r.set("t", "ready_ack").set("pid", (int64_t)g_agent.pid);
-    ck::send_frame(cl.fd, r);
+    send_to_client(cl, r);
   } else if (t == "agent_initialized") {
     size_t slash = g_marker.rfind('/');
     if (slash != std::string::npos) ck::mkdirs(g_marker.substr(0, slash));
     ck::write_file(g_marker, std::to_string(time(nullptr)));
     mj::Value r; r.set("t", "ok");
-    ck::send_frame(cl.fd, r);
+    send_to_client(cl, r);
   } else if (t == "exec") {
     start_exec(cl, req);
   } else if (t == "attach") {
     cl.attached = true;
     mj::Value r; r.set("t", "attached").set("tty", g_agent.tty);
-    ck::send_frame(cl.fd, r);
+    send_to_client(cl, r);
   } else if (t == "stdin") {
     auto data = ck::b64_decode(req["data"].as_str());
     if (g_agent.in >= 0 && !data.empty())
@@ -534,10 +584,10 @@ void handle_frame(Client& cl, const mj::Value& req) {
     r.set("t", "status").set("cmd_running", g_agent.running)
      .set("exit_code", (int64_t)g_agent.exit_code)
      .set("initialized", ck::exists(g_marker));
-    ck::send_frame(cl.fd, r);
+    send_to_client(cl, r);
   } else {
     mj::Value r; r.set("t", "error").set("msg", "unknown command: " + t);
-    ck::send_frame(cl.fd, r);
+    send_to_client(cl, r);
   }
 }
 
@@ -559,7 +609,7 @@ void drain_client(Client& cl, bool& drop) {
           handle_frame(cl, mj::parse(body));
         } catch (const std::exception& e) {
           mj::Value r; r.set("t", "error").set("msg", e.what());
-          ck::send_frame(cl.fd, r);
+          send_to_client(cl, r);
         }
       }
       continue;
@@ -625,7 +675,11 @@ int main() {
     pfds.push_back({g_selfpipe[0], POLLIN, 0});
     pfds.push_back({listen_fd, POLLIN, 0});
     size_t client_base = pfds.size();
-    for (auto& c : g_clients) pfds.push_back({c.fd, POLLIN, 0});
+    for (auto& c : g_clients) {
+      short ev = POLLIN;
+      if (!c.out.empty()) ev |= POLLOUT;
+      pfds.push_back({c.fd, ev, 0});
+    }
     int agent_io_idx = -1, agent_err_idx = -1;
     if (g_agent.io >= 0) { agent_io_idx = pfds.size(); pfds.push_back({g_agent.io, POLLIN, 0}); }
     if (g_agent.err >= 0) { agent_err_idx = pfds.size(); pfds.push_back({g_agent.err, POLLIN, 0}); }
@@ -658,11 +712,13 @@ int main() {
     for (size_t i = 0; i < g_clients.size(); i++) {
       if (client_base + i >= pfds.size()) break;
       auto& pe = pfds[client_base + i];
+      if (pe.revents & POLLOUT) flush_client(g_clients[i]);
       if (pe.revents & (POLLIN | POLLHUP | POLLERR)) {
         bool drop = false;
         drain_client(g_clients[i], drop);
         if (drop) drop_fds.push_back(g_clients[i].fd);
       }
+      if (g_clients[i].dead) drop_fds.push_back(g_clients[i].fd);
     }
     for (int fd : drop_fds) {
       for (size_t i = 0; i < g_clients.size(); i++) {
