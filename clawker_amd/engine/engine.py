@@ -282,6 +282,11 @@ class Engine:
         status = self._status(rundir)
         if status.get("state") == "running" and self._pid_alive(status.get("pid")):
             raise ConflictError(f"sandbox already running: {name}")
+        # a dying previous instance's shim may still be reaping; let it
+        # finish so its exit.json/status.json writes can't clobber ours
+        old_shim = status.get("shim_pid")
+        if old_shim and self._pid_alive(int(old_shim)):
+            self._wait_pid_gone(int(old_shim), 5.0)
         # clear stale run state (incl. a dead instance's control socket —
         # start() readiness keys on its existence)
         for f in ("exit.json", "status.json", "pid", "console.log",

@@ -44,7 +44,11 @@ def compile_policy(bypass: bool = False,
 
 
 def write_policy_snapshot(rundir: Path, policy: dict) -> None:
-    tmp = rundir / ".policy.json.tmp"
+    # unique temp name: the CP watcher and explicit attach calls may write
+    # concurrently; a shared temp name loses the atomic-rename race
+    import os
+    import threading
+    tmp = rundir / f".policy.{os.getpid()}.{threading.get_ident()}.tmp"
     tmp.write_text(json.dumps(policy, indent=1))
     tmp.replace(rundir / "policy.json")
 
