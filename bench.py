@@ -87,7 +87,8 @@ def main() -> int:
     ws = Path(os.environ["CLAWKER_STATE_DIR"]) / "workspace"
     ws.mkdir(parents=True, exist_ok=True)
     (ws / ".clawker.yaml").write_text(
-        "project: bench\nworkspace:\n  share_volume: false\n")
+        "project: bench\nagent:\n  harness: echo\n"
+        "workspace:\n  share_volume: false\n")
     (ws / "task.txt").write_text("synthetic agent task\n")
 
     from clawker_amd.config import load_config
