@@ -116,6 +116,7 @@ def exporter_running() -> bool:
 def ensure_running(port: int) -> None:
     if exporter_running():
         return
+    consts.log_dir().mkdir(parents=True, exist_ok=True)
     logf = open(consts.log_dir() / "exporter.out", "ab")
     subprocess.Popen(
         [sys.executable, "-m", "clawker_amd.monitor.exporter", str(port)],

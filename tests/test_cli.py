@@ -5,6 +5,7 @@ import json
 import os
 import subprocess
 import sys
+import time
 from pathlib import Path
 
 import pytest
@@ -156,3 +157,53 @@ def test_worktree_cli(proj):
     r = _invoke(["worktree", "remove", "feature/x"])
     assert r.exit_code == 0
     assert json.loads(_invoke(["worktree", "list", "--format", "json"]).output) == []
+
+
+@requires_isolation
+def test_cli_cp_copy_roundtrip(proj):
+    """clawker cp: host->sandbox and sandbox->host tar streaming."""
+    _invoke(["init", "--yes", "--name", "cptest", "--harness", "echo"])
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+
+    def clawker(*args, timeout=60):
+        return subprocess.run([sys.executable, "-m", "clawker_amd", *args],
+                              capture_output=True, text=True, timeout=timeout,
+                              cwd=str(proj), env=env)
+
+    r = clawker("run", "-d", "--agent", "c1", "--no-firewall",
+                "--", "/bin/sh", "-c", "echo sandbox-data > /srcfile; sleep 30")
+    assert r.returncode == 0, r.stderr
+    time.sleep(0.3)
+    # sandbox -> host
+    r = clawker("cp", "clawker.cptest.c1:/srcfile", str(proj / "out"))
+    assert r.returncode == 0, r.stderr
+    assert (proj / "out" / "srcfile").read_text().strip() == "sandbox-data"
+    # host -> sandbox
+    (proj / "payload.txt").write_text("host-data")
+    r = clawker("cp", str(proj / "payload.txt"), "c1:/incoming")
+    assert r.returncode == 0, r.stderr
+    r = clawker("exec", "c1", "--", "/bin/cat", "/incoming/payload.txt")
+    assert "host-data" in r.stdout
+    clawker("rm", "-f", "c1")
+
+
+def test_doctor_reports_capabilities(isolated_env):
+    r = _invoke(["doctor", "--format", "json"])
+    checks = {c["check"]: c for c in json.loads(r.output)}
+    assert checks["root"]["ok"]
+    assert "native runtime" in checks
+    assert checks["openssl"]["ok"]
+
+
+def test_monitor_exporter_metrics(isolated_env):
+    import urllib.request
+    from clawker_amd.monitor import exporter
+    port = 19193
+    exporter.ensure_running(port)
+    try:
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/metrics", timeout=5).read().decode()
+        assert "clawker_sandboxes_running" in body
+        assert "clawker_gpu_busy_percent" in body   # HELP/TYPE always present
+    finally:
+        exporter.stop_running()
