@@ -53,8 +53,21 @@ def cp_agents(ctx: Ctx):
 
 @cp_group.command("events")
 @click.option("-n", type=int, default=50, show_default=True)
+@click.option("-f", "--follow", is_flag=True)
 @pass_factory
-def cp_events(ctx: Ctx, n):
+def cp_events(ctx: Ctx, n, follow):
     cp = ctx.factory.controlplane()
+    seen = 0
     for ev in cp.events(n):
         ctx.factory.io.print(json.dumps(ev))
+        seen = max(seen, int(ev.get("ts", 0) * 1e6))
+    if not follow:
+        return
+    import time as _t
+    while True:
+        _t.sleep(1.0)
+        for ev in cp.events(200):
+            ts = int(ev.get("ts", 0) * 1e6)
+            if ts > seen:
+                ctx.factory.io.print(json.dumps(ev))
+                seen = max(seen, ts)

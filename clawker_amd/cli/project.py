@@ -47,6 +47,20 @@ def init_cmd(ctx: Ctx, name, preset, harness, gpus, vcs, yes):
     cfg_path = root / consts.PROJECT_FILE_NAME
     if cfg_path.exists() or (root / consts.PROJECT_DIR_NAME).exists():
         raise ClawkerError(f"project already initialized at {root}")
+    # inside an existing project's subdirectory: create a walk-up override
+    # layer instead of a new registration (reference: init.go behavior)
+    cfg = f.config()
+    if cfg.project_root is not None and cfg.project_root != root:
+        doc = {}
+        if harness != "claude":
+            doc["agent"] = {"harness": harness}
+        if gpus:
+            doc["gpu"] = {"count": gpus}
+        cfg_path.write_text(yaml.safe_dump(doc, sort_keys=False) if doc else "{}\n")
+        f.io.success(
+            f"created override layer {cfg_path} inside project "
+            f"'{cfg.project_slug}' (root: {cfg.project_root}); not registered")
+        return
     slug = slugify(name or root.name)
 
     if not yes and f.io.can_prompt():
@@ -142,3 +156,24 @@ def project_info(ctx: Ctx):
 def project_remove(ctx: Ctx, name):
     ProjectRegistry().unregister(name)
     ctx.factory.io.success(f"unregistered '{name}'")
+
+
+@project_group.command("edit")
+@pass_factory
+def project_edit(ctx: Ctx):
+    """Open the project's clawker.yaml in $EDITOR (validated on save)."""
+    import os
+    import subprocess
+    f = ctx.factory
+    cfg = f.config(require_project=True)
+    path = None
+    for layer in cfg.project_store.layers:
+        if layer.path is not None:
+            path = layer.path
+    if path is None:
+        raise ClawkerError("no project config file found")
+    editor = os.environ.get("EDITOR") or os.environ.get("VISUAL") or "nano"
+    subprocess.run([editor, str(path)], check=False)
+    from ..config import load_config
+    load_config(f.cwd).project   # validate (raises on schema errors)
+    f.io.success(f"saved {path}")

@@ -90,9 +90,8 @@ def load_config(cwd: Path | None = None, require_project: bool = False) -> Confi
     cwd = (cwd or Path.cwd()).resolve()
     paths = discover_project_layers(cwd)
     layers: list[Layer] = []
-    root: Path | None = None
+    roots: list[Path] = []
     for p in paths:
-        # layer name = file identity; nearest project file defines the root
         name = "project-local" if p.name == consts.PROJECT_LOCAL_BASENAME else "project"
         if name == "project":
             name = f"project:{p.parent}"
@@ -100,7 +99,23 @@ def load_config(cwd: Path | None = None, require_project: bool = False) -> Confi
         base = p.parent
         if base.name == consts.PROJECT_DIR_NAME:
             base = base.parent
-        root = base
+        if base not in roots:
+            roots.append(base)
+    # the project ROOT is registry-resolved when possible (reference:
+    # project identity resolution, resolve.go); otherwise the OUTERMOST
+    # discovered config dir — nested config files inside a project are
+    # walk-up override layers, not new roots
+    root: Path | None = None
+    if roots:
+        try:
+            from ..project import ProjectRegistry
+            entry = ProjectRegistry().resolve_by_path(cwd)
+        except Exception:
+            entry = None
+        if entry is not None and Path(entry.root) in roots:
+            root = Path(entry.root)
+        else:
+            root = roots[0]     # outermost (discover order: farthest first)
     if require_project and root is None:
         raise ClawkerError(
             "no clawker project found (run `clawker init` at your project root)")

@@ -207,3 +207,26 @@ def test_monitor_exporter_metrics(isolated_env):
         assert "clawker_gpu_busy_percent" in body   # HELP/TYPE always present
     finally:
         exporter.stop_running()
+
+
+def test_init_override_layer_in_subdir(proj):
+    """init inside an existing project creates a walk-up override, not a
+    new registration (reference: init.go subdirectory behavior)."""
+    _invoke(["init", "--yes", "--name", "outer", "--harness", "echo"])
+    sub = proj / "nested" / "dir"
+    sub.mkdir(parents=True)
+    os.chdir(sub)
+    try:
+        r = _invoke(["init", "--yes", "--harness", "codex"])
+        assert r.exit_code == 0, r.output
+        assert "override layer" in r.output
+        assert (sub / ".clawker.yaml").exists()
+        from clawker_amd.config import load_config
+        cfg = load_config(sub)
+        assert cfg.project_root == proj          # still the outer project
+        assert cfg.project_slug == "outer"
+        assert cfg.project.agent.harness == "codex"   # override applies
+        from clawker_amd.project import ProjectRegistry
+        assert len(ProjectRegistry().list_projects()) == 1
+    finally:
+        os.chdir(proj)
