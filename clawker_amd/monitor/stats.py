@@ -41,12 +41,26 @@ class StatsSnapshot:
     sandboxes: list[dict] = field(default_factory=list)
     gpus: list[GpuSample] = field(default_factory=list)
     allocations: dict = field(default_factory=dict)   # gpu index -> sandbox
+    events: list[dict] = field(default_factory=list)  # recent egress decisions
 
 
 _sampler: RocmSampler | None = None
 
 
-def collect_stats(engine: Engine) -> StatsSnapshot:
+def _console_tail(rundir: Path, max_len: int = 72) -> str:
+    try:
+        with open(rundir / "console.log", "rb") as f:
+            f.seek(0, 2)
+            size = f.tell()
+            f.seek(max(0, size - 4096))
+            lines = [l for l in f.read().decode(errors="replace").splitlines()
+                     if l.strip()]
+            return lines[-1][-max_len:] if lines else ""
+    except OSError:
+        return ""
+
+
+def collect_stats(engine: Engine, with_tails: bool = True) -> StatsSnapshot:
     global _sampler
     snap = StatsSnapshot()
     for info in engine.list():
@@ -56,6 +70,8 @@ def collect_stats(engine: Engine) -> StatsSnapshot:
         }
         if info.state == "running":
             row.update(cgroup_stats(info.name))
+        if with_tails:
+            row["tail"] = _console_tail(info.rundir)
         snap.sandboxes.append(row)
         for g in info.gpus:
             snap.allocations[g] = info.name
@@ -65,6 +81,12 @@ def collect_stats(engine: Engine) -> StatsSnapshot:
         snap.gpus = _sampler.sample()
     except Exception:
         snap.gpus = []
+    try:
+        from ..controlplane.daemon import EventLog, events_path
+        snap.events = [e for e in EventLog(events_path()).tail(40)
+                       if e.get("event") == "egress_decision"][-8:]
+    except Exception:
+        snap.events = []
     return snap
 
 

@@ -30,16 +30,31 @@ def render_dashboard(snap: StatsSnapshot):
     panes.append(Panel(gt, title="MI355X GPUs", border_style="cyan"))
 
     st = Table(box=None, expand=True, pad_edge=False)
-    for c in ("SANDBOX", "STATE", "MEM", "PIDS", "GPUS"):
+    for c in ("SANDBOX", "STATE", "MEM", "GPUS", "CONSOLE"):
         st.add_column(c)
     for s in snap.sandboxes:
         mem = s.get("mem_bytes")
         state_style = {"running": "green", "exited": "dim"}.get(s["state"], "yellow")
         st.add_row(s["name"], f"[{state_style}]{s['state']}[/{state_style}]",
                    f"{mem / 2**20:.0f}M" if mem else "-",
-                   str(s.get("pids") or "-"),
-                   ",".join(map(str, s["gpus"])) or "-")
+                   ",".join(map(str, s["gpus"])) or "-",
+                   f"[dim]{s.get('tail', '')}[/dim]")
     panes.append(Panel(st, title="agent sandboxes", border_style="magenta"))
+
+    if snap.events:
+        et = Table(box=None, expand=True, pad_edge=False)
+        for c in ("TIME", "SANDBOX", "ACTION", "DST"):
+            et.add_column(c)
+        for e in snap.events:
+            act = e.get("action", "")
+            style = "green" if act in ("allow", "resolve") else "red"
+            dst = e.get("dst", "")
+            if e.get("path"):
+                dst += e["path"]
+            et.add_row(time.strftime("%H:%M:%S", time.localtime(e.get("ts", 0))),
+                       str(e.get("sandbox", ""))[-24:],
+                       f"[{style}]{act}[/{style}]", dst[:48])
+        panes.append(Panel(et, title="egress decisions", border_style="yellow"))
     panes.append(f"[dim]{time.strftime('%H:%M:%S', time.localtime(snap.ts))} — "
                  f"ctrl-c to exit[/dim]")
     return Group(*panes)

@@ -75,3 +75,33 @@ def test_fleet_snapshot_mode_without_git(isolated_env, tmp_path):
     assert (root / "seed.txt").read_text() == "s"   # snapshots, not binds
     fleet.down()
     fleet.orch.close()
+
+
+def test_fleet_8way_worktrees_with_firewall(git_proj):
+    """BASELINE config 3 shape: 8 concurrent worktree sandboxes with the
+    egress firewall enabled (netns + gateways), one agent loop each."""
+    from clawker_amd.config import load_config
+    from clawker_amd.fleet import Fleet, FleetOptions
+    cfg = load_config(git_proj)
+    fleet = Fleet(cfg)
+    members = fleet.up(FleetOptions(
+        count=8, firewall=True, branch_prefix="w",
+        cmd=["/bin/sh", "-c",
+             "python3 -c \"import socket,sys\n"
+             "s=socket.socket(); s.settimeout(2)\n"
+             "try: s.connect(('203.0.113.9',443)); sys.exit(1)\n"
+             "except OSError: sys.exit(0)\" && echo FW_OK $CLAWKER_AGENT"]))
+    assert len(members) == 8
+    fleet.wait(members, timeout_s=120)
+    failures = [(m.sandbox, m.exit_code) for m in members if m.exit_code != 0]
+    assert not failures, failures
+    for i, m in enumerate(members):
+        logs = fleet.orch.engine.logs(m.sandbox).decode()
+        assert f"FW_OK w{i}" in logs
+        # gateway sockets were attached for every member
+        rundir = fleet.orch.engine.inspect(m.sandbox).rundir
+        assert (rundir / "policy.json").exists()
+    fleet.down(branch_prefix="w")
+    from clawker_amd.controlplane.client import CPClient
+    CPClient(auto_start=False).stop()
+    fleet.orch.close()
