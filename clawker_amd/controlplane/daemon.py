@@ -130,6 +130,10 @@ class CPDaemon:
         self._bypass_timer: threading.Timer | None = None
         self._registry_db = self._open_registry()
         self._last_agent_seen = time.time()
+        # sandboxes explicitly firewall-disabled by the admin: the watcher
+        # must not re-enroll them (reference: FirewallDisable sticks until
+        # FirewallEnable, CLAUDE.md:69 semantics)
+        self._fw_disabled: set[str] = set()
 
     # ------------------------------------------------------------ registry --
     def _open_registry(self) -> sqlite3.Connection:
@@ -165,7 +169,8 @@ class CPDaemon:
                 # reconcile firewall gateways against live state (reference:
                 # dockerevents reconcile + FirewallEnable drift guard)
                 live_fw = {i.name: i for i in running
-                           if i.labels.get("dev.clawker.fw") == "on"}
+                           if i.labels.get("dev.clawker.fw") == "on"
+                           and i.name not in self._fw_disabled}
                 for name, i in live_fw.items():
                     if name not in self.gateways.gateways:
                         self._attach_gateway(name, i.rundir)
@@ -260,9 +265,12 @@ class CPDaemon:
             self._set_bypass(int(req.get("seconds", 0)))
             return {"ok": True, "until": self._bypass_until}
         if op == "fw_attach":
+            self._fw_disabled.discard(req["sandbox"])
             self._attach_gateway(req["sandbox"], req["rundir"])
             return {"ok": True}
         if op == "fw_detach":
+            if req.get("sticky"):
+                self._fw_disabled.add(req["sandbox"])
             self.gateways.detach(req["sandbox"])
             return {"ok": True}
         if op == "events":

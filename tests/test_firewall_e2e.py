@@ -223,3 +223,26 @@ print("KEEPALIVE " + json.dumps(out), flush=True)
     assert res[1] == ["/secret/x", 403]     # mid-stream request DENIED
     assert res[2] == ["/ok2", 200]          # session still usable
     orch.teardown(name, force=True)
+
+
+def test_disable_is_sticky_against_watcher(fw_env):
+    """firewall disable must not be undone by the CP watcher's reconcile
+    (reference: FirewallDisable sticks until FirewallEnable)."""
+    orch, ws, port = fw_env
+    from clawker_amd.controlplane.client import CPClient
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.fwtest.st"
+    orch.run(RunOptions(agent="st", name=name, autostart=True, firewall=True,
+                        cmd=["sleep", "30"]))
+    assert _wait_gateway(orch, name)
+    rundir = orch.engine.inspect(name).rundir
+    cp = CPClient()
+    cp.request({"op": "fw_detach", "sandbox": name, "sticky": True})
+    (rundir / "egress.sock").unlink(missing_ok=True)
+    # give the watcher 2+ reconcile cycles: it must NOT re-attach
+    time.sleep(2.5)
+    assert not (rundir / "egress.sock").exists()
+    # explicit re-enable works
+    cp.request({"op": "fw_attach", "sandbox": name, "rundir": str(rundir)})
+    assert (rundir / "egress.sock").exists()
+    orch.teardown(name, force=True)

@@ -105,3 +105,37 @@ def test_harness_config_volume_persists_across_recreate(isolated_env, tmp_path):
     finally:
         orch.teardown(name, force=True)
         orch.close()
+
+
+@requires_isolation
+def test_real_claude_harness_runs_in_sandbox(isolated_env, tmp_path):
+    """The CI image ships the actual Claude Code CLI: run it (--version)
+    inside a firewalled sandbox with the claude harness wiring (config
+    volume, managed env, egress floor project)."""
+    import shutil as _sh
+    if not _sh.which("claude"):
+        pytest.skip("claude CLI not in image")
+    ws = tmp_path / "clproj"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text("project: cltest\n")   # harness: claude default
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    orch = Orchestrator(load_config(ws))
+    name = "clawker.cltest.a"
+    try:
+        orch.run(RunOptions(agent="a", name=name, autostart=True, firewall=True,
+                            user="root", cmd=["claude", "--version"]))
+        code = orch.engine.wait(name, timeout_s=60)
+        logs = orch.engine.logs(name).decode()
+        assert code == 0, logs
+        assert "Claude Code" in logs or any(c.isdigit() for c in logs), logs
+        # harness config volume mounted
+        info = orch.engine.inspect(name)
+        import json as _json
+        spec = _json.loads((info.rundir / "spec.json").read_text())
+        assert any(m["dst"].endswith("/.claude") for m in spec["mounts"])
+    finally:
+        orch.teardown(name, force=True)
+        from clawker_amd.controlplane.client import CPClient
+        CPClient(auto_start=False).stop()
+        orch.close()
