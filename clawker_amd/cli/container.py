@@ -12,7 +12,6 @@ import time
 
 import click
 
-from .. import consts
 from ..cmdutil import Factory, format_age, resolve_sandbox_name
 from ..errors import ClawkerError, ExitError
 from ..orchestrator import RunOptions

@@ -128,7 +128,6 @@ def stack_group():
 @stack_group.command("list")
 @pass_factory
 def stack_list(ctx: Ctx):
-    from pathlib import Path
     from ..bundle.loader import ASSETS, load_stack
     f = ctx.factory
     names = sorted(p.stem for p in (ASSETS / "stacks").glob("*.yaml"))

@@ -3,7 +3,6 @@ project init first-run UX)."""
 from __future__ import annotations
 
 import json
-from pathlib import Path
 
 import click
 import yaml

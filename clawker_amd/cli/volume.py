@@ -8,7 +8,6 @@ import time
 import click
 
 from ..cmdutil import format_age
-from ..errors import ClawkerError
 from .root import Ctx, cli, pass_factory
 
 

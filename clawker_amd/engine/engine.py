@@ -22,7 +22,7 @@ from ..errors import ConflictError, EngineError, NotFoundError
 from ..logger import get as get_logger
 from .ckd_client import CkdClient
 from .images import HOSTFS, ImageStore
-from .spec import Device, Mount, SandboxSpec
+from .spec import Mount, SandboxSpec
 from .state import StateDB
 
 log = get_logger("engine")

@@ -17,7 +17,7 @@ from .config import Config
 from .engine import CkdClient, Engine, SandboxSpec
 from .engine.images import HOSTFS
 from .engine.spec import Device, Mount
-from .errors import ClawkerError, ConflictError
+from .errors import ClawkerError
 from .gpu import GPUAllocator
 from .logger import get as get_logger
 
