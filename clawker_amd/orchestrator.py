@@ -287,7 +287,13 @@ class Orchestrator:
         return self.engine.create(spec, image=image)
 
     def start(self, name: str):
-        return self.engine.start(name)
+        """engine start + firewall gateway enrollment (reference:
+        ContainerStart then FirewallEnable, container_start.go:349) —
+        every start path (CLI run/start, fleet, bench) goes through here."""
+        info = self.engine.start(name)
+        if info.labels.get("dev.clawker.fw") == "on":
+            self._fw_attach(info)
+        return info
 
     def run(self, opts: RunOptions):
         """create + start (+ firewall gateway attach); returns SandboxInfo
@@ -295,20 +301,34 @@ class Orchestrator:
         agent_ready)."""
         info = self.create(opts)
         try:
-            started = self.engine.start(info.name)
-            if info.labels.get("dev.clawker.fw") == "on":
-                self._fw_attach(info)
-            return started
+            return self.start(info.name)
         except BaseException:
             self.teardown(info.name, force=True)
             raise
 
     def _fw_attach(self, info) -> None:
-        """Bind the host-side policy gateway for this sandbox via the CP
-        (reference: FirewallEnable enrolling the container post-start,
-        container_start.go:349)."""
+        """Push the composed egress rules (harness floor ∪ project rules ∪
+        add_domains) and bind the host-side policy gateway via the CP
+        (reference: FirewallInit + FirewallAddRules in
+        BootstrapServicesPreStart, then FirewallEnable post-start)."""
         from .controlplane.client import CPClient
+        from .storage import to_plain
         cp = CPClient()
+        proj = self.cfg.project
+        rules = list(proj.security.egress)
+        from .config.schema import EgressRule
+        for d in proj.security.add_domains:
+            rules.append(EgressRule(dst=d, proto="tls", port=443))
+        try:
+            from .bundle import load_harness
+            from .bundle.loader import harness_egress_floor
+            harness = load_harness(proj.agent.harness, self.cfg.project_root)
+            rules = harness_egress_floor(harness, rules)
+        except ClawkerError:
+            pass
+        if rules:
+            cp.request({"op": "fw_add_rules",
+                        "rules": [to_plain(r) for r in rules]})
         cp.request({"op": "fw_attach", "sandbox": info.name,
                     "rundir": str(info.rundir)})
 
