@@ -46,6 +46,8 @@ def _run_opts_flags(fn):
     fn = click.option("--label", "-l", "labels_kv", multiple=True, help="KEY=VALUE label")(fn)
     fn = click.option("--no-host-services", is_flag=True,
                       help="skip hostproxy + ssh/gpg agent bridges")(fn)
+    fn = click.option("--restart", "restart_policy", default="no",
+                      metavar="no|on-failure[:N]", show_default=True)(fn)
     return fn
 
 
@@ -82,7 +84,7 @@ def _build_opts(f: Factory, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
         workspace_mode=workspace_mode, firewall=firewall,
         mem_bytes=_parse_mem(mem), pids_max=pids_limit,
         labels=_parse_kv(labels_kv), autostart=autostart,
-        host_services=not no_host_services)
+        host_services=not no_host_services, restart=restart_policy)
     if worktree:
         from ..project.worktrees import ensure_worktree
         wt = ensure_worktree(f.config(require_project=True), worktree)
@@ -135,7 +137,7 @@ def _boot_and_wait(f: Factory, name: str, interactive: bool, tty: bool,
 @pass_factory
 def run_cmd(ctx: Ctx, agent, image, gpus, gpu_indices, hbm_gb, env_kv, workdir,
             user, workspace_mode, worktree, firewall, mem, pids_limit,
-            labels_kv, no_host_services, interactive, tty, detach, rm_after, cmd):
+            labels_kv, no_host_services, restart_policy, interactive, tty, detach, rm_after, cmd):
     """Create and start an agent sandbox (alias of `container run`).
 
     CMD may start with an image reference: `clawker run @ -- <cmd>` runs the
@@ -152,6 +154,7 @@ def run_cmd(ctx: Ctx, agent, image, gpus, gpu_indices, hbm_gb, env_kv, workdir,
                        workdir, user, workspace_mode, worktree, firewall, mem,
                        pids_limit, labels_kv, cmd, tty, autostart=False,
                        no_host_services=no_host_services)
+    opts.restart = restart_policy
     orch = f.orchestrator()
     info = orch.create(opts)
     code = _boot_and_wait(f, info.name, interactive, tty, detach, rm_after)
@@ -169,13 +172,14 @@ container_group.add_command(run_cmd, "run")
 @pass_factory
 def create_cmd(ctx: Ctx, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
                workdir, user, workspace_mode, worktree, firewall, mem,
-               pids_limit, labels_kv, no_host_services, tty, cmd):
+               pids_limit, labels_kv, no_host_services, restart_policy, tty, cmd):
     """Create a sandbox without starting it."""
     f = ctx.factory
     opts = _build_opts(f, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
                        workdir, user, workspace_mode, worktree, firewall, mem,
                        pids_limit, labels_kv, list(cmd), tty, autostart=False,
                        no_host_services=no_host_services)
+    opts.restart = restart_policy
     info = f.orchestrator().create(opts)
     f.io.print(info.name)
 

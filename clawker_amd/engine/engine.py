@@ -359,12 +359,19 @@ class Engine:
             pid = self._init_pid(rundir)
             if pid is not None and not self._pid_alive(pid):
                 # shim writes exit.json right after reaping; brief grace
+                # (a restart-policy shim may instead respawn — detect via a
+                # changed pidfile and keep waiting)
                 for _ in range(100):
                     code = self._exit_code(rundir)
                     if code is not None:
                         return code
+                    new_pid = self._init_pid(rundir)
+                    if new_pid not in (None, pid) and self._pid_alive(new_pid):
+                        break   # restarted; continue outer wait
                     time.sleep(0.01)
-                return -1
+                else:
+                    return -1
+                continue
             if deadline is not None and time.monotonic() > deadline:
                 raise EngineError("wait", f"timeout waiting for {name}")
             time.sleep(0.005)

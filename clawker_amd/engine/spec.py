@@ -52,6 +52,9 @@ class SandboxSpec:
     paths: dict[str, str] = field(default_factory=dict)
     # auxiliary in-sandbox daemons ckd supervises (e.g. ckgw gateway shims)
     services: list[dict] = field(default_factory=list)
+    # restart policy (reference: docker-style on-failure:N)
+    restart_policy: str = "no"       # no | on-failure
+    restart_max: int = 3
 
     def to_json(self) -> str:
         d = {
@@ -60,6 +63,7 @@ class SandboxSpec:
             "rundir": self.rundir,
             "paths": self.paths,
             "services": self.services,
+            "restart": {"policy": self.restart_policy, "max": self.restart_max},
             "rootfs": {
                 "lowerdirs": self.lowerdirs,
                 "upper": self.upper,

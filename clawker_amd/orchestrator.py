@@ -45,6 +45,7 @@ class RunOptions:
     mounts: list[Mount] = field(default_factory=list)
     labels: dict = field(default_factory=dict)
     name: str = ""                   # override computed sandbox name
+    restart: str = "no"              # no | on-failure[:N]
     # hostproxy + ssh/gpg agent bridges; the CLI enables this by default,
     # programmatic callers (bench, tests) opt in explicitly
     host_services: bool = False
@@ -267,10 +268,13 @@ class Orchestrator:
             if host_ssh_auth_sock():
                 env.setdefault("SSH_AUTH_SOCK", "/run/clawker/ssh-agent.sock")
 
+        restart_policy, _, restart_n = opts.restart.partition(":")
         spec = SandboxSpec(
             name=name,
             hostname=f"{self.cfg.project_slug or 'clawker'}-{opts.agent}"[:63],
             services=services,
+            restart_policy=restart_policy or "no",
+            restart_max=int(restart_n) if restart_n.isdigit() else 3,
             netns=effective_firewall,
             tty=opts.tty,
             autostart=opts.autostart,
@@ -289,7 +293,12 @@ class Orchestrator:
     def start(self, name: str):
         """engine start + firewall gateway enrollment (reference:
         ContainerStart then FirewallEnable, container_start.go:349) —
-        every start path (CLI run/start, fleet, bench) goes through here."""
+        every start path (CLI run/start, fleet, bench) goes through here.
+
+        Note: with autostart=True the agent CMD may race the gateway
+        attach by a few ms; the window is FAIL-CLOSED (no gateway = all
+        egress refused). The CLI/fleet paths use autostart=False and
+        release the CMD only after this returns."""
         info = self.engine.start(name)
         if info.labels.get("dev.clawker.fw") == "on":
             self._fw_attach(info)

@@ -73,6 +73,8 @@ struct Spec {
   int64_t mem_bytes = 0;
   int64_t pids_max = 0;
   bool device_allow_only = true;
+  std::string restart = "no";  // no | on-failure
+  int64_t restart_max = 3;
   mj::Value raw;               // full spec for ckd
 };
 
@@ -113,6 +115,11 @@ Spec parse_spec(const std::string& path) {
   s.mem_bytes = cg["mem_bytes"].as_int(0);
   s.pids_max = cg["pids"].as_int(0);
   s.device_allow_only = cg["device_allow_only"].as_bool(true);
+  if (v.has("restart")) {
+    s.restart = v["restart"]["policy"].as_str();
+    if (s.restart.empty()) s.restart = "no";
+    s.restart_max = v["restart"]["max"].as_int(3);
+  }
   if (s.name.empty() || s.rundir.empty())
     die("spec: name/rundir required");
   if (s.backend == "ns" && (s.merged.empty() || s.lowerdirs.empty()))
@@ -444,6 +451,8 @@ int run(const std::string& spec_path) {
   cgroups_setup(s);
   trace("cgroups_setup");
 
+  int attempts = 0;
+restart_attempt:
   if (pipe2(g_sync_pipe, O_CLOEXEC) != 0) die("pipe");
 
   if (s.backend == "proc") {
@@ -496,6 +505,20 @@ int run(const std::string& spec_path) {
   int code = 0, sig = 0;
   if (WIFEXITED(wstatus)) code = WEXITSTATUS(wstatus);
   else if (WIFSIGNALED(wstatus)) { sig = WTERMSIG(wstatus); code = 128 + sig; }
+
+  // restart policy (reference: docker-style on-failure:N — the CP
+  // container itself runs with on-failure:3, bootstrap.go:501)
+  if (s.restart == "on-failure" && code != 0 && attempts < s.restart_max) {
+    attempts++;
+    warn("restarting sandbox (exit %d, attempt %d/%lld)", code, attempts,
+         (long long)s.restart_max);
+    mj::Value st;
+    st.set("state", "restarting").set("attempt", (int64_t)attempts);
+    ck::write_file(s.rundir + "/status.json", st.dump());
+    unlink((s.rundir + "/" + "ctl.sock").c_str());
+    usleep(200000 * attempts);   // linear backoff
+    goto restart_attempt;
+  }
 
   {
     mj::Value ex;

@@ -145,3 +145,30 @@ def test_stale_state_cleared_on_restart_after_kill(orch):
     orch.engine.start(name)
     assert orch.engine.inspect(name).state == "running"
     orch.engine.stop(name)
+
+
+@requires_isolation
+def test_restart_policy_on_failure(orch):
+    """on-failure:N restart policy (reference: the CP's on-failure:3)."""
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.rtest.rp"
+    # fails twice (marker counts attempts), succeeds on the third
+    opts = RunOptions(
+        agent="rp", name=name, autostart=True, restart="on-failure:3",
+        cmd=["/bin/sh", "-c",
+             "n=$(cat /run/clawker/attempts 2>/dev/null || echo 0); "
+             "n=$((n+1)); echo $n > /run/clawker/attempts; "
+             "echo attempt-$n; [ $n -ge 3 ] && exit 0 || exit 7"])
+    orch.run(opts)
+    code = orch.engine.wait(name, timeout_s=60)
+    logs = orch.engine.logs(name).decode()
+    assert code == 0, logs
+    assert "attempt-1" in logs and "attempt-3" in logs
+    orch.teardown(name, force=True)
+
+    # exhausted retries: final failure code propagates
+    name2 = "clawker.rtest.rp2"
+    orch.run(RunOptions(agent="rp2", name=name2, autostart=True,
+                        restart="on-failure:2", cmd=["/bin/sh", "-c", "exit 5"]))
+    assert orch.engine.wait(name2, timeout_s=60) == 5
+    orch.teardown(name2, force=True)
