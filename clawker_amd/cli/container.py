@@ -84,7 +84,7 @@ def _build_opts(f: Factory, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
         workspace_mode=workspace_mode, firewall=firewall,
         mem_bytes=_parse_mem(mem), pids_max=pids_limit,
         labels=_parse_kv(labels_kv), autostart=autostart,
-        host_services=not no_host_services, restart=restart_policy)
+        host_services=not no_host_services)
     if worktree:
         from ..project.worktrees import ensure_worktree
         wt = ensure_worktree(f.config(require_project=True), worktree)
