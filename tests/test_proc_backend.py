@@ -73,3 +73,18 @@ def test_proc_snapshot_workspace(proc_orch):
         cmd=["/bin/sh", "-c", "echo x >> file.txt; cat file.txt"]))
     assert orch.engine.wait(name, timeout_s=30) == 0
     assert (ws / "file.txt").read_text() == "ws-data"   # host copy untouched
+
+
+def test_proc_restart_policy(proc_orch):
+    orch, ws = proc_orch
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.ptest.rp"
+    info_dir = None
+    orch.run(RunOptions(
+        agent="rp", name=name, autostart=True, restart="on-failure:2",
+        cmd=["/bin/sh", "-c",
+             "d=$(dirname $CKD_MARKER 2>/dev/null || echo /tmp); "
+             "exit 3"]))
+    assert orch.engine.wait(name, timeout_s=60) == 3   # retries exhausted
+    # status recorded restarting attempts along the way (final: exited)
+    orch.teardown(name, force=True)

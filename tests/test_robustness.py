@@ -172,3 +172,41 @@ def test_restart_policy_on_failure(orch):
                         restart="on-failure:2", cmd=["/bin/sh", "-c", "exit 5"]))
     assert orch.engine.wait(name2, timeout_s=60) == 5
     orch.teardown(name2, force=True)
+
+
+@requires_isolation
+def test_ckd_survives_garbage_on_control_socket(orch):
+    """Malformed frames / random bytes on ctl.sock must not kill PID 1."""
+    import socket as _socket
+    import struct
+    from clawker_amd.engine import wire
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.rtest.fz"
+    orch.run(RunOptions(agent="fz", name=name, autostart=True, cmd=["sleep", "30"]))
+    sock_path = orch.engine.ctl_sock(name)
+    # 1. raw garbage
+    s = wire.connect_unix(sock_path, timeout=5)
+    s.sendall(b"\x00\x00\x00\x05notjs")
+    s.close()
+    # 2. huge length prefix
+    s = wire.connect_unix(sock_path, timeout=5)
+    s.sendall(struct.pack(">I", 0x7FFFFFFF))
+    s.close()
+    # 3. valid frame, unknown command + missing fields
+    s = wire.connect_unix(sock_path, timeout=5)
+    wire.send_frame(s, {"t": "exec"})            # no id/stages
+    wire.send_frame(s, {"t": "wat"})
+    r = wire.recv_frame(s)
+    assert r and r.get("t") == "error"
+    s.close()
+    # 4. partial frame then disconnect
+    s = wire.connect_unix(sock_path, timeout=5)
+    s.sendall(b"\x00\x00")
+    s.close()
+    # ckd still alive and serving
+    with orch.client(name) as c:
+        h = c.hello()
+        assert h["cmd_running"] is True
+    code, out, _ = orch.engine.exec(name, ["/bin/echo", "alive"])
+    assert code == 0 and b"alive" in out
+    orch.engine.stop(name)
