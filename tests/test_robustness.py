@@ -210,3 +210,22 @@ def test_ckd_survives_garbage_on_control_socket(orch):
     code, out, _ = orch.engine.exec(name, ["/bin/echo", "alive"])
     assert code == 0 and b"alive" in out
     orch.engine.stop(name)
+
+
+@requires_isolation
+def test_audit_events_written(orch):
+    """ckd writes the load-bearing audit lines (reference: clawkerd
+    session/shell_command audit contract)."""
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.rtest.au"
+    orch.run(RunOptions(agent="au", name=name, autostart=True,
+                        cmd=["/bin/sh", "-c", "sleep 2; exit 4"]))
+    orch.engine.exec(name, ["/bin/echo", "x"])
+    orch.engine.wait(name, timeout_s=30)
+    audit = (orch.engine.inspect(name).rundir / "audit.jsonl").read_text()
+    events = [json.loads(l)["event"] for l in audit.splitlines()]
+    assert "agent_spawned" in events
+    assert "session_started" in events
+    assert "shell_command_started" in events
+    assert "shell_command_done" in events
+    assert "agent_exit" in events
