@@ -76,6 +76,9 @@ def image_rm(ctx: Ctx, names):
         f.io.print(n)
 
 
+cli.add_command(image_rm, "rmi")   # docker-style top-level alias
+
+
 @image_group.command("prune")
 @pass_factory
 def image_prune(ctx: Ctx):
