@@ -71,3 +71,12 @@ def cp_events(ctx: Ctx, n, follow):
             if ts > seen:
                 ctx.factory.io.print(json.dumps(ev))
                 seen = max(seen, ts)
+
+
+@cp_group.command("serve", hidden=True)
+def cp_serve():
+    """Run the control-plane daemon in the foreground (hidden; normally
+    spawned on demand — reference: hidden hostproxy/bridge daemons)."""
+    import sys
+    from ..controlplane.daemon import main as cpd_main
+    sys.exit(cpd_main())

@@ -134,3 +134,21 @@ def dashboard_cmd(ctx: Ctx, interval):
                 time.sleep(interval)
         except KeyboardInterrupt:
             pass
+
+
+@monitor_group.command("serve", hidden=True)
+@click.option("--port", type=int, default=19090)
+def monitor_serve(port):
+    """Run the metrics exporter in the foreground (hidden)."""
+    import sys
+    from ..monitor.exporter import serve
+    sys.exit(serve(port))
+
+
+@cli.command("hostproxy", hidden=True)
+def hostproxy_serve():
+    """Run the hostproxy daemon in the foreground (hidden; reference:
+    `clawker host-proxy serve`)."""
+    import sys
+    from ..hostproxy.daemon import main as hp_main
+    sys.exit(hp_main())

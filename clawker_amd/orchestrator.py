@@ -194,6 +194,17 @@ class Orchestrator:
                 env["CLAWKER_HBM_GB"] = str(hbm)
                 pct = max(1, min(100, round(hbm * 100 / 288)))
                 env["GPU_MAX_ALLOC_PERCENT"] = str(pct)
+        # remaining A.1 contract: editor default, terminal capability
+        # passthrough, telemetry segmentation attributes
+        import os as _os
+        env.setdefault("EDITOR", _os.environ.get("EDITOR", "nano"))
+        if opts.tty:
+            if _os.environ.get("TERM"):
+                env.setdefault("TERM", _os.environ["TERM"])
+            if _os.environ.get("COLORTERM"):
+                env.setdefault("COLORTERM", _os.environ["COLORTERM"])
+        env.setdefault("OTEL_RESOURCE_ATTRIBUTES",
+                       f"project={self.cfg.project_slug},agent={opts.agent}")
         if proj.agent.env_file:
             from .dotenv import parse_env_file
             base_dir = self.cfg.project_root or Path.cwd()
@@ -209,6 +220,9 @@ class Orchestrator:
 
         cmd = opts.cmd or proj.agent.cmd
         user = opts.user if opts.user is not None else ""
+        if not user and harness is not None:
+            user = harness.user
+        env.setdefault(consts.ENV_USER, user or "root")
         default_workdir = str(ws_effective) if ws_src is not None else "/"
         workdir = opts.workdir or proj.agent.workdir or default_workdir
 
