@@ -130,3 +130,52 @@ def test_discovery_nearest_wins(tmp_path):
     found = discover_project_layers(inner)
     # nearest last = highest priority
     assert found[-1].parent == inner
+
+
+def test_invalid_yaml_top_level_rejected(tmp_path):
+    from clawker_amd.storage.store import StoreError
+    p = tmp_path / "bad.yaml"
+    p.write_text("- just\n- a\n- list\n")
+    with pytest.raises(StoreError):
+        Store(SchemaT, [Layer("main", p)])
+    with pytest.raises(StoreError):
+        Store.from_string(SchemaT, "[1, 2]")
+
+
+def test_remove_and_get_path(tmp_path):
+    s = Store.from_string(SchemaT, "title: x\ninner:\n  x: 5\n")
+    assert s.get_path("inner.x") == 5
+    assert s.get_path("inner.missing", "dflt") == "dflt"
+    assert s.remove("inner.x") is True
+    assert s.remove("inner.x") is False
+    assert s.get().inner.x == 1   # default shows through again
+
+
+def test_write_layer_virtual_errors():
+    from clawker_amd.storage.store import StoreError
+    s = Store.from_string(SchemaT, "{}")
+    with pytest.raises(StoreError):
+        s.write_layer("defaults")
+    with pytest.raises(StoreError):
+        s.set("title", "x", layer="nope")
+
+
+def test_migration_chain_applies_in_order(tmp_path):
+    p = tmp_path / "m.yaml"
+    p.write_text("version: 0\ntitle: a\n")
+    calls = []
+
+    def m1(d):
+        calls.append(1)
+        d["title"] += "-1"
+        return d
+
+    def m2(d):
+        calls.append(2)
+        d["title"] += "-2"
+        return d
+
+    s = Store(SchemaT, [Layer("main", p)], migrations=[(2, m2), (1, m1)])
+    assert calls == [1, 2]          # sorted by version despite declaration order
+    assert s.get().title == "a-1-2"
+    assert yaml.safe_load(p.read_text())["version"] == 2
