@@ -553,3 +553,42 @@ def restart_cmd(ctx: Ctx, timeout, names):
 
 
 container_group.add_command(restart_cmd, "restart")
+
+
+@container_group.command("update")
+@click.option("--memory", "-m", "mem", default="", help="new memory limit (e.g. 8g)")
+@click.option("--pids-limit", type=int, default=0)
+@click.argument("name")
+@pass_factory
+def update_cmd(ctx: Ctx, mem, pids_limit, name):
+    """Update resource limits of a RUNNING sandbox (live cgroup rewrite)."""
+    from pathlib import Path as _P
+    f = ctx.factory
+    sb = resolve_sandbox_name(f, name)
+    info = f.engine().inspect(sb)
+    if info.state != "running":
+        raise ClawkerError(f"sandbox not running: {sb}")
+    cg_root = _P("/sys/fs/cgroup")
+    changed = []
+    if mem:
+        nbytes = _parse_mem(mem)
+        for p in (cg_root / "memory" / "clawker" / sb / "memory.limit_in_bytes",
+                  cg_root / "clawker" / sb / "memory.max"):
+            if p.parent.is_dir():
+                p.write_text(str(nbytes))
+                changed.append(f"memory={mem}")
+                break
+        else:
+            raise ClawkerError(
+                "no memory cgroup for this sandbox (started without -m; "
+                "restart with --memory to enable live updates)")
+    if pids_limit:
+        for p in (cg_root / "pids" / "clawker" / sb / "pids.max",
+                  cg_root / "clawker" / sb / "pids.max"):
+            if p.parent.is_dir():
+                p.write_text(str(pids_limit))
+                changed.append(f"pids={pids_limit}")
+                break
+    if not changed:
+        raise ClawkerError("nothing to update (pass --memory/--pids-limit)")
+    f.io.success(f"updated {sb}: {', '.join(changed)}")

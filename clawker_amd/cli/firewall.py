@@ -80,16 +80,23 @@ def fw_status(ctx: Ctx):
     f = ctx.factory
     rules = _store().list()
     infos = f.engine().list()
-    enforced = [i.name for i in infos if i.state == "running"
-                and i.labels.get("dev.clawker.fw") != "off"]
     backend = f.engine().backend
-    f.io.print(json.dumps({
+    out = {
         "backend": backend,
-        "enforcement": "netns+gateway" if backend == "ns" else
+        "enforcement": "netns+gateway+mitm" if backend == "ns" else
                        "unavailable (proc backend: no netns on this host)",
         "rules": len(rules),
         "running_sandboxes": len([i for i in infos if i.state == "running"]),
-    }, indent=1))
+    }
+    try:
+        cp = f.controlplane()
+        cp.auto_start = False
+        if cp.running():
+            out.update(cp.request({"op": "fw_status"}))
+            out.pop("ok", None)
+    except Exception:
+        pass
+    f.io.print(json.dumps(out, indent=1))
 
 
 @firewall_group.command("reload")
