@@ -24,6 +24,42 @@ from .logger import get as get_logger
 log = get_logger("orchestrator")
 
 
+def _clawkerignore_filter(ws_root: Path):
+    """shutil.copytree ignore callable from .clawkerignore (reference:
+    .clawkerignore excluding paths from snapshot workspaces). Patterns are
+    fnmatch-style against names and workspace-relative paths; a trailing
+    '/' means directories only; '#' comments."""
+    import fnmatch
+    patterns: list[tuple[str, bool]] = []
+    try:
+        for line in (ws_root / consts.IGNORE_FILE_NAME).read_text().splitlines():
+            line = line.strip()
+            if not line or line.startswith("#"):
+                continue
+            dir_only = line.endswith("/")
+            patterns.append((line.rstrip("/"), dir_only))
+    except OSError:
+        pass
+    if not patterns:
+        return None
+
+    def ignore(dirpath, names):
+        rel_dir = Path(dirpath).resolve().relative_to(ws_root.resolve())
+        out = set()
+        for n in names:
+            rel = str(rel_dir / n) if str(rel_dir) != "." else n
+            is_dir = (Path(dirpath) / n).is_dir()
+            for pat, dir_only in patterns:
+                if dir_only and not is_dir:
+                    continue
+                if fnmatch.fnmatch(n, pat) or fnmatch.fnmatch(rel, pat):
+                    out.add(n)
+                    break
+        return out
+
+    return ignore
+
+
 @dataclass
 class RunOptions:
     agent: str = "agent"
@@ -115,7 +151,9 @@ class Orchestrator:
                 vol_path, fresh = self.engine.ensure_volume(
                     vol_name, {consts.MANAGED_LABEL: "true"})
                 if fresh:
-                    shutil.copytree(ws_src, vol_path, dirs_exist_ok=True, symlinks=True)
+                    shutil.copytree(ws_src, vol_path, dirs_exist_ok=True,
+                                    symlinks=True,
+                                    ignore=_clawkerignore_filter(Path(ws_src)))
                 mounts.append(Mount(src=str(vol_path), dst=ws_dst))
                 ws_effective = vol_path if backend == "proc" else Path(ws_dst)
             else:

@@ -121,3 +121,24 @@ def test_bench_distributed_gloo_world2():
     out = json.loads(line)
     assert out["n_gpus"] == 2
     assert out["config"]["concurrent_loops"] == 2
+
+
+@requires_isolation
+def test_clawkerignore_respected_in_snapshots(orch, proj):
+    from clawker_amd.orchestrator import RunOptions
+    (proj / ".clawkerignore").write_text("# comment\n.git/\n*.secret\nbuild/\n")
+    (proj / ".git").mkdir()
+    (proj / ".git" / "HEAD").write_text("ref: x")
+    (proj / "api.secret").write_text("k")
+    (proj / "build").mkdir()
+    (proj / "build" / "out.o").write_text("o")
+    (proj / "keep.txt").write_text("keep")
+    name = "clawker.otest.ign"
+    orch.run(RunOptions(agent="ign", name=name, autostart=True,
+                        workspace_mode="snapshot", firewall=False,
+                        cmd=["/bin/sh", "-c", "ls -a /workspace"]))
+    assert orch.engine.wait(name, timeout_s=30) == 0
+    out = orch.engine.logs(name).decode()
+    assert "keep.txt" in out and "data.txt" in out
+    assert ".git" not in out and "api.secret" not in out and "build" not in out
+    orch.teardown(name, force=True)
