@@ -34,6 +34,9 @@ class Harness:
     name: str = ""
     description: str = ""
     cmd: list[str] = field(default_factory=list)
+    # how to hand a one-shot prompt to the harness; "@PROMPT_FILE@" is
+    # replaced with the in-sandbox prompt path (fleet --prompt)
+    prompt_cmd: list[str] = field(default_factory=list)
     user: str = "agent"
     stacks: list[str] = field(default_factory=list, metadata={"merge": "union"})
     install: list[BuildStep] = field(default_factory=list)

@@ -23,11 +23,13 @@ def fleet_group():
 @click.option("--base", default="", help="worktree start point")
 @click.option("--no-worktrees", is_flag=True, help="snapshot workspaces instead")
 @click.option("--firewall/--no-firewall", "firewall", default=None)
+@click.option("--prompt", "prompt_file", default="",
+              help="one-shot prompt file handed to each agent (harness prompt_cmd)")
 @click.option("--watch", is_flag=True, help="open the live dashboard after start")
 @click.argument("cmd", nargs=-1, type=click.UNPROCESSED)
 @pass_factory
 def fleet_up(ctx: Ctx, count, branch_prefix, gpus_per_agent, image, base,
-             no_worktrees, firewall, watch, cmd):
+             no_worktrees, firewall, prompt_file, watch, cmd):
     """Start N agent sandboxes over worktrees with 1:1 GPU pinning."""
     f = ctx.factory
     cfg = f.config(require_project=True)
@@ -35,6 +37,7 @@ def fleet_up(ctx: Ctx, count, branch_prefix, gpus_per_agent, image, base,
     members = fleet.up(FleetOptions(
         count=count, branch_prefix=branch_prefix, gpus_per_agent=gpus_per_agent,
         cmd=list(cmd), image=image, firewall=firewall, base=base,
+        prompt_file=prompt_file,
         use_worktrees=False if no_worktrees else None))
     for m in members:
         gpus = ",".join(map(str, m.gpus)) or "-"

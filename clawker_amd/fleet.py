@@ -32,6 +32,7 @@ class FleetOptions:
     firewall: bool | None = None
     use_worktrees: bool | None = None    # None => auto (git repo present)
     base: str = ""                       # worktree start point
+    prompt_file: str = ""                # one-shot prompt handed to each agent
 
 
 @dataclass
@@ -57,10 +58,21 @@ class Fleet:
             use_wt = (self.cfg.project_root is not None
                       and is_git_repo(self.cfg.project_root))
         members: list[FleetMember] = []
+        prompt_cmd: list[str] = []
+        if opts.prompt_file and not opts.cmd:
+            from .bundle import load_harness
+            harness = load_harness(self.cfg.project.agent.harness,
+                                   self.cfg.project_root)
+            if not harness.prompt_cmd:
+                raise ClawkerError(
+                    f"harness '{harness.name}' declares no prompt_cmd")
+            prompt_cmd = [a.replace("@PROMPT_FILE@", "/run/clawker/prompt.md")
+                          for a in harness.prompt_cmd]
         for i in range(opts.count):
             agent = f"{opts.branch_prefix}{i}"
             ropts = RunOptions(
-                agent=agent, cmd=list(opts.cmd), env=dict(opts.env),
+                agent=agent, cmd=list(opts.cmd) or list(prompt_cmd),
+                env=dict(opts.env),
                 image=opts.image, gpus=opts.gpus_per_agent,
                 firewall=opts.firewall, autostart=False)
             branch = ""
@@ -71,7 +83,11 @@ class Fleet:
             else:
                 # disposable copies so parallel agents never collide
                 ropts.workspace_mode = "snapshot"
-            info = self.orch.run(ropts)
+            info = self.orch.create(ropts)
+            if opts.prompt_file:
+                import shutil as _sh
+                _sh.copy2(opts.prompt_file, info.rundir / "prompt.md")
+            self.orch.start(info.name)
             # drive init/boot plans then release the CMD (reference: the CP
             # Executor path — fan-out members get the same boot contract)
             from .controlplane.plans import drive_boot

@@ -105,3 +105,25 @@ def test_fleet_8way_worktrees_with_firewall(git_proj):
     from clawker_amd.controlplane.client import CPClient
     CPClient(auto_start=False).stop()
     fleet.orch.close()
+
+
+def test_fleet_prompt_file(isolated_env, tmp_path):
+    """fleet --prompt: each agent gets the prompt via the harness's
+    prompt_cmd (the autonomous-loop driver shape, BASELINE config 4)."""
+    root = tmp_path / "pfproj"
+    root.mkdir()
+    (root / ".clawker.yaml").write_text("project: pftest\nagent:\n  harness: echo\n")
+    (root / "task.md").write_text("refactor the widget\n")
+    from clawker_amd.config import load_config
+    from clawker_amd.fleet import Fleet, FleetOptions
+    fleet = Fleet(load_config(root))
+    members = fleet.up(FleetOptions(
+        count=2, firewall=False, use_worktrees=False,
+        prompt_file=str(root / "task.md")))
+    fleet.wait(members, timeout_s=60)
+    for m in members:
+        assert m.exit_code == 0
+        logs = fleet.orch.engine.logs(m.sandbox).decode()
+        assert "PROMPT:" in logs and "refactor the widget" in logs
+    fleet.down()
+    fleet.orch.close()
