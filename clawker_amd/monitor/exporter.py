@@ -62,6 +62,17 @@ def _metrics_text() -> str:
         if s.get("pids"):
             lines.append(f"clawker_sandbox_pids{{{lbl}}} {s['pids']}")
     lines.append(f"clawker_sandboxes_running {running}")
+    # plugin lane: merge third-party textfile metrics (the reference's
+    # monitoring-units collector routing, reduced to the node-exporter
+    # textfile pattern: drop .prom files into <state>/metrics.d/)
+    metrics_d = consts.state_dir() / "metrics.d"
+    if metrics_d.is_dir():
+        for p in sorted(metrics_d.glob("*.prom")):
+            try:
+                lines.append(f"# collector: {p.name}")
+                lines.append(p.read_text().strip())
+            except OSError:
+                continue
     return "\n".join(lines) + "\n"
 
 

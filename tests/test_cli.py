@@ -205,6 +205,14 @@ def test_monitor_exporter_metrics(isolated_env):
             f"http://127.0.0.1:{port}/metrics", timeout=5).read().decode()
         assert "clawker_sandboxes_running" in body
         assert "clawker_gpu_busy_percent" in body   # HELP/TYPE always present
+        # plugin textfile lane (monitoring-units analog)
+        from clawker_amd import consts
+        md = consts.state_dir() / "metrics.d"
+        md.mkdir(parents=True, exist_ok=True)
+        (md / "myunit.prom").write_text("myunit_metric 42\n")
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/metrics", timeout=5).read().decode()
+        assert "myunit_metric 42" in body
     finally:
         exporter.stop_running()
 
