@@ -20,18 +20,36 @@ def _read_int(p: Path) -> int | None:
         return None
 
 
+_cpu_prev: dict[str, tuple[float, int]] = {}   # name -> (ts, cpu_ns)
+
+
 def cgroup_stats(name: str) -> dict:
-    """Memory + pids for a sandbox's cgroup (v1 hybrid or v2 unified)."""
+    """Memory + pids + CPU%% for a sandbox's cgroup (v1 hybrid or v2)."""
     out: dict = {}
+    cpu_ns = None
     v2 = _CG_ROOT / "clawker" / name
     if (v2 / "memory.current").exists():
         out["mem_bytes"] = _read_int(v2 / "memory.current")
         out["pids"] = _read_int(v2 / "pids.current")
-        return out
-    mem = _CG_ROOT / "memory" / "clawker" / name / "memory.usage_in_bytes"
-    pids = _CG_ROOT / "pids" / "clawker" / name / "pids.current"
-    out["mem_bytes"] = _read_int(mem)
-    out["pids"] = _read_int(pids)
+        try:
+            for line in (v2 / "cpu.stat").read_text().splitlines():
+                if line.startswith("usage_usec"):
+                    cpu_ns = int(line.split()[1]) * 1000
+        except OSError:
+            pass
+    else:
+        out["mem_bytes"] = _read_int(
+            _CG_ROOT / "memory" / "clawker" / name / "memory.usage_in_bytes")
+        out["pids"] = _read_int(
+            _CG_ROOT / "pids" / "clawker" / name / "pids.current")
+        cpu_ns = _read_int(_CG_ROOT / "cpuacct" / "clawker" / name / "cpuacct.usage")
+    if cpu_ns is not None:
+        now = time.monotonic()
+        prev = _cpu_prev.get(name)
+        _cpu_prev[name] = (now, cpu_ns)
+        if prev and now > prev[0]:
+            out["cpu_pct"] = round(
+                100.0 * (cpu_ns - prev[1]) / ((now - prev[0]) * 1e9), 1)
     return out
 
 
@@ -94,11 +112,13 @@ def render_stats(snap: StatsSnapshot):
     from rich.console import Group
     from rich.table import Table
     t = Table(title="sandboxes", box=None, pad_edge=False)
-    for c in ("NAME", "STATE", "PID", "MEM", "PIDS", "GPUS"):
+    for c in ("NAME", "STATE", "PID", "CPU", "MEM", "PIDS", "GPUS"):
         t.add_column(c)
     for s in snap.sandboxes:
         mem = s.get("mem_bytes")
+        cpu = s.get("cpu_pct")
         t.add_row(s["name"], s["state"], str(s.get("pid") or "-"),
+                  f"{cpu:.0f}%" if cpu is not None else "-",
                   f"{mem / 2**20:.0f}M" if mem else "-",
                   str(s.get("pids") or "-"),
                   ",".join(map(str, s["gpus"])) or "-")
