@@ -320,16 +320,25 @@ container_group.add_command(ps_cmd, "ps")
 @click.option("-u", "--user", default="", help="run as user")
 @click.option("-w", "--workdir", default="")
 @click.option("-e", "--env", "env_kv", multiple=True)
+@click.option("-i", "--interactive", is_flag=True, help="keep stdin open")
+@click.option("-t", "--tty", is_flag=True, help="allocate a pseudo-TTY")
 @click.argument("name")
 @click.argument("cmd", nargs=-1, required=True, type=click.UNPROCESSED)
 @pass_factory
-def exec_cmd(ctx: Ctx, user, workdir, env_kv, name, cmd):
-    """Run a command in a running sandbox."""
+def exec_cmd(ctx: Ctx, user, workdir, env_kv, interactive, tty, name, cmd):
+    """Run a command in a running sandbox (-it for an interactive shell)."""
     f = ctx.factory
     sb = resolve_sandbox_name(f, name)
+    if tty:
+        code = _exec_interactive(f, sb, list(cmd), user, workdir,
+                                 _parse_kv(env_kv) or None)
+        if code != 0:
+            raise ExitError(code)
+        return
     stdin = b""
-    if not sys.stdin.isatty():
-        stdin = sys.stdin.buffer.read()
+    if interactive or not sys.stdin.isatty():
+        if not sys.stdin.isatty():
+            stdin = sys.stdin.buffer.read()
     with f.engine().client(sb) as c:
         stage = {"argv": list(cmd)}
         if user:
@@ -341,6 +350,62 @@ def exec_cmd(ctx: Ctx, user, workdir, env_kv, name, cmd):
     sys.stderr.buffer.write(err)
     if code != 0:
         raise ExitError(code)
+
+
+def _exec_interactive(f: Factory, sb: str, argv, user, workdir, env) -> int:
+    """Interactive pty exec: raw-mode pump like attach, but frame-scoped
+    to the exec id."""
+    import os
+    import select
+    import termios
+    import tty as _tty
+    from ..engine import wire as _wire
+    with f.engine().client(sb) as c:
+        eid = c.exec_start_tty(argv, user=user, cwd=workdir, env=env)
+        stdin_fd = sys.stdin.fileno()
+        saved = None
+        if sys.stdin.isatty():
+            saved = termios.tcgetattr(stdin_fd)
+            _tty.setraw(stdin_fd)
+            try:
+                sz = os.get_terminal_size(sys.stdout.fileno())
+                c.exec_resize(eid, sz.lines, sz.columns)
+            except OSError:
+                pass
+        code = -1
+        try:
+            sock = c.sock
+            sock.settimeout(None)
+            while True:
+                rlist = [sock] + ([stdin_fd] if saved is not None or interactive_stdin() else [])
+                ready, _, _ = select.select(rlist, [], [])
+                if stdin_fd in ready:
+                    data = os.read(stdin_fd, 65536)
+                    if not data:
+                        break
+                    c.exec_stdin(eid, data)
+                if sock in ready:
+                    fr = _wire.recv_frame(sock)
+                    if fr is None:
+                        break
+                    t = fr.get("t")
+                    if t == "out" and fr.get("id") == eid:
+                        sys.stdout.buffer.write(_wire.unb64(fr.get("data", "")))
+                        sys.stdout.buffer.flush()
+                    elif t == "done" and fr.get("id") == eid:
+                        code = int(fr.get("code", -1))
+                        break
+        finally:
+            if saved is not None:
+                termios.tcsetattr(stdin_fd, termios.TCSADRAIN, saved)
+        return code
+
+
+def interactive_stdin() -> bool:
+    try:
+        return not sys.stdin.closed
+    except Exception:
+        return False
 
 
 container_group.add_command(exec_cmd, "exec")

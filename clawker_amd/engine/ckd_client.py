@@ -151,6 +151,32 @@ class CkdClient:
             # ignore unrelated events
         return code, bytes(out), bytes(err)
 
+    def exec_start_tty(self, argv: list[str], user: str = "", cwd: str = "",
+                       env: dict[str, str] | None = None) -> str:
+        """Start an interactive (pty) exec; returns its id. The caller
+        pumps frames: out(id) frames arrive, exec_stdin/exec_resize go
+        back, done(id) carries the exit code."""
+        CkdClient._exec_seq += 1
+        eid = f"t{CkdClient._exec_seq}-{int(time.time() * 1000) & 0xFFFFFF}"
+        stage: dict[str, Any] = {"argv": argv}
+        if user:
+            stage["user"] = user
+        if cwd:
+            stage["cwd"] = cwd
+        msg: dict[str, Any] = {"t": "exec", "id": eid, "tty": True,
+                               "stages": [stage]}
+        if env:
+            msg["env"] = env
+        self.send(msg)
+        self._wait_for("started")
+        return eid
+
+    def exec_stdin(self, eid: str, data: bytes) -> None:
+        self.send({"t": "exec_stdin", "id": eid, "data": wire.b64(data)})
+
+    def exec_resize(self, eid: str, rows: int, cols: int) -> None:
+        self.send({"t": "exec_resize", "id": eid, "rows": rows, "cols": cols})
+
     # -- console streaming ---------------------------------------------------
     def stream_events(self) -> Iterator[dict[str, Any]]:
         """Yield frames until the connection closes (use after attach())."""
