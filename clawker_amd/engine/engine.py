@@ -279,6 +279,19 @@ class Engine:
     def start(self, name: str, wait_ready_s: float = 10.0) -> SandboxInfo:
         row = self._row(name)
         rundir = Path(row["rundir"])
+        # serialize concurrent starts of the same sandbox (two CLIs racing
+        # would otherwise both pass the running check and spawn two shims)
+        import fcntl as _fcntl
+        lock_f = open(rundir / "start.lock", "w")
+        try:
+            _fcntl.flock(lock_f, _fcntl.LOCK_EX)
+            return self._start_locked(name, row, rundir, wait_ready_s)
+        finally:
+            _fcntl.flock(lock_f, _fcntl.LOCK_UN)
+            lock_f.close()
+
+    def _start_locked(self, name: str, row: dict, rundir: Path,
+                      wait_ready_s: float) -> SandboxInfo:
         status = self._status(rundir)
         if status.get("state") == "running" and self._pid_alive(status.get("pid")):
             raise ConflictError(f"sandbox already running: {name}")

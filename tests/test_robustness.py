@@ -229,3 +229,67 @@ def test_audit_events_written(orch):
     assert "shell_command_started" in events
     assert "shell_command_done" in events
     assert "agent_exit" in events
+
+
+@requires_isolation
+def test_concurrent_start_race_single_instance(orch):
+    """Two racing starts yield exactly one running instance."""
+    from clawker_amd.errors import ConflictError
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.rtest.race"
+    opts = RunOptions(agent="race", name=name, cmd=["sleep", "20"])
+    opts.autostart = True
+    orch.create(opts)
+    results = []
+
+    def starter():
+        try:
+            orch.engine.start(name)
+            results.append("started")
+        except ConflictError:
+            results.append("conflict")
+        except Exception as e:
+            results.append(f"error:{e}")
+
+    ts = [threading.Thread(target=starter) for _ in range(3)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=30)
+    assert results.count("started") == 1, results
+    assert all(r in ("started", "conflict") for r in results), results
+    # exactly one ckd instance
+    code, out, _ = orch.engine.exec(name, ["/bin/sh", "-c", "echo one"])
+    assert code == 0
+    orch.engine.stop(name)
+
+
+@requires_isolation
+def test_multiple_attach_clients_see_console(orch):
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.rtest.ma"
+    opts = RunOptions(agent="ma", name=name,
+                      cmd=["/bin/sh", "-c", "sleep 0.3; echo SHARED-LINE; sleep 1"])
+    opts.autostart = False
+    orch.run(opts)
+    c1 = orch.client(name)
+    c2 = orch.client(name)
+    try:
+        c1.attach()
+        c2.attach()
+        c1.agent_ready()
+        seen = []
+        for c in (c1, c2):
+            data = b""
+            for fr in c.stream_events():
+                if fr.get("t") == "console":
+                    from clawker_amd.engine import wire
+                    data += wire.unb64(fr.get("data", ""))
+                if b"SHARED-LINE" in data or fr.get("t") == "agent_exit":
+                    break
+            seen.append(b"SHARED-LINE" in data)
+        assert seen == [True, True]
+    finally:
+        c1.close()
+        c2.close()
+        orch.engine.stop(name)
