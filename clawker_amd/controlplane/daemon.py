@@ -180,8 +180,10 @@ class CPDaemon:
 
     # ------------------------------------------------------------- watcher --
     def _watch_loop(self) -> None:
-        """Track sandbox lifecycle; drive plans for detach-started
-        sandboxes; drain-to-zero."""
+        """Track sandbox lifecycle: agent registry, firewall-gateway and
+        agent-socket-bridge reconcile, drain-to-zero. (Init/boot plans are
+        driven by the starting client — CLI/fleet — which holds the ckd
+        session; the watcher only reconciles host-side attachments.)"""
         while not self._stop.is_set():
             try:
                 infos = self.engine.list()
