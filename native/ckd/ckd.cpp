@@ -155,7 +155,7 @@ std::vector<char*> to_argv(const std::vector<std::string>& v) {
 // ------------------------------------------------------------- clients -----
 
 struct Client {
-  int fd;
+  int fd = -1;
   std::string buf;          // partial inbound frame bytes
   std::string out;          // pending outbound bytes (nonblocking writes)
   bool attached = false;    // subscribed to console stream
@@ -606,7 +606,6 @@ void pump_console(int fd, int stream) {
 
 // ------------------------------------------------------------- handlers ----
 
-bool g_ready_announced = false;
 
 void handle_frame(Client& cl, const mj::Value& req) {
   const std::string& t = req["t"].as_str();
@@ -798,7 +797,9 @@ int main() {
     if (pfds[1].revents & POLLIN) {
       int cfd;
       while ((cfd = accept4(listen_fd, nullptr, nullptr, SOCK_NONBLOCK | SOCK_CLOEXEC)) >= 0) {
-        g_clients.push_back(Client{cfd});
+        Client nc;
+        nc.fd = cfd;
+        g_clients.push_back(std::move(nc));
         audit("session_started", {{"clients", mj::Value((int64_t)g_clients.size())}});
       }
     }

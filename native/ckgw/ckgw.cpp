@@ -104,7 +104,10 @@ int run_tcp(const char* ip, int port, const char* sock_path) {
         fcntl(ufd, F_SETFL, O_NONBLOCK);
         int one = 1;
         setsockopt(cfd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
-        relays.push_back(Relay{cfd, ufd});
+        Relay nr;
+        nr.a = cfd;
+        nr.b = ufd;
+        relays.push_back(std::move(nr));
       }
     }
     size_t pi = 1;

@@ -350,10 +350,10 @@ int child_main(void*) {
   ck::mkdirs(m + "/dev/shm");
   mnt("tmpfs", (m + "/dev/shm").c_str(), "tmpfs", MS_NOSUID | MS_NODEV,
       "mode=1777,size=65536k");
-  symlink("/proc/self/fd", (m + "/dev/fd").c_str());
-  symlink("/proc/self/fd/0", (m + "/dev/stdin").c_str());
-  symlink("/proc/self/fd/1", (m + "/dev/stdout").c_str());
-  symlink("/proc/self/fd/2", (m + "/dev/stderr").c_str());
+  if (symlink("/proc/self/fd", (m + "/dev/fd").c_str()) != 0) warn("symlink");
+  if (symlink("/proc/self/fd/0", (m + "/dev/stdin").c_str()) != 0) warn("symlink");
+  if (symlink("/proc/self/fd/1", (m + "/dev/stdout").c_str()) != 0) warn("symlink");
+  if (symlink("/proc/self/fd/2", (m + "/dev/stderr").c_str()) != 0) warn("symlink");
 
   // writable scratch
   ck::mkdirs(m + "/tmp");
