@@ -46,9 +46,16 @@ def settings_set(ctx: Ctx, path, value):
 
 
 @settings_group.command("edit")
+@click.option("-i", "--interactive", is_flag=True,
+              help="field-by-field editor instead of $EDITOR")
 @pass_factory
-def settings_edit(ctx: Ctx):
-    """Open settings.yaml in $EDITOR."""
+def settings_edit(ctx: Ctx, interactive):
+    """Edit settings.yaml ($EDITOR, or -i for the field editor)."""
+    if interactive:
+        from ..storeui import edit_store
+        n = edit_store(load_settings(), ctx.factory.io)
+        ctx.factory.io.success(f"{n} field(s) updated")
+        return
     path = consts.config_dir() / consts.SETTINGS_BASENAME
     path.parent.mkdir(parents=True, exist_ok=True)
     if not path.exists():

@@ -59,9 +59,16 @@ class EventLog:
         self.path.parent.mkdir(parents=True, exist_ok=True)
         self._lock = threading.Lock()
 
+    MAX_BYTES = 10 * 1024 * 1024
+
     def emit(self, event: str, **kv) -> None:
         rec = {"ts": time.time(), "event": event, **kv}
         with self._lock:
+            try:
+                if self.path.stat().st_size > self.MAX_BYTES:
+                    self.path.replace(self.path.with_suffix(".jsonl.1"))
+            except OSError:
+                pass
             with open(self.path, "a") as f:
                 f.write(json.dumps(rec) + "\n")
 
