@@ -88,6 +88,8 @@ class IOStreams:
 class TestIOStreams(IOStreams):
     """Buffer trio for tests (reference: iostreams.Test())."""
 
+    __test__ = False    # not a pytest collection target
+
     def __init__(self, stdin_text: str = ""):
         self.in_buf = io.StringIO(stdin_text)
         self.out_buf = io.StringIO()
