@@ -67,12 +67,13 @@ def image_inspect(ctx: Ctx, names):
 
 
 @image_group.command("rm")
+@click.option("-f", "--force", is_flag=True, help="remove even if in use")
 @click.argument("names", nargs=-1, required=True)
 @pass_factory
-def image_rm(ctx: Ctx, names):
+def image_rm(ctx: Ctx, force, names):
     f = ctx.factory
     for n in names:
-        f.engine().images.remove(n)
+        f.engine().remove_image(n, force=force)
         f.io.print(n)
 
 

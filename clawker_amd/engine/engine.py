@@ -457,6 +457,18 @@ class Engine:
                 stage["cwd"] = cwd
             return c.exec([stage], stdin=stdin)
 
+    # ------------------------------------------------------------- images ---
+    def remove_image(self, name: str, force: bool = False) -> None:
+        """Remove an image, refusing while any sandbox still references it
+        (a running sandbox's overlay must not lose its lowerdirs —
+        docker-rmi semantics)."""
+        users = [row["name"] for row in self.db.list_sandboxes()
+                 if row["image"] == name]
+        if users and not force:
+            raise ConflictError(
+                f"image in use by sandboxes: {users} (remove them or --force)")
+        self.images.remove(name)
+
     # ------------------------------------------------------------ volumes ---
     def ensure_volume(self, name: str, labels: dict | None = None) -> tuple[Path, bool]:
         """Create-or-get a named volume; returns (path, fresh)."""

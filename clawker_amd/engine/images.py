@@ -187,13 +187,23 @@ class ImageStore:
         return dirs
 
     def prune_layers(self) -> int:
-        """Remove layers referenced by no image."""
+        """Remove layers referenced by no image. Staging layers (tmp-*) of
+        a possibly in-flight build are only pruned once stale (>2h) — a
+        concurrent build's upper dir must never vanish underneath it."""
         referenced: set[str] = set()
         for meta in self.list():
             referenced.update(meta.layers)
         removed = 0
+        now = time.time()
         for d in (self.root / "layers").iterdir():
-            if d.name not in referenced:
-                shutil.rmtree(d, ignore_errors=True)
-                removed += 1
+            if d.name in referenced:
+                continue
+            if d.name.startswith("tmp-"):
+                try:
+                    if now - d.stat().st_mtime < 2 * 3600:
+                        continue
+                except OSError:
+                    continue
+            shutil.rmtree(d, ignore_errors=True)
+            removed += 1
         return removed

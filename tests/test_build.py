@@ -110,3 +110,31 @@ def test_boot_plans_execute_from_image(ctx):
     assert "post-init hook executed" in out
     assert "pre-run hook executed" in out
     orch.teardown(sb, force=True)
+
+
+@requires_isolation
+def test_image_rm_refuses_while_in_use(ctx):
+    cfg, eng = ctx
+    from clawker_amd.bundler import Builder
+    from clawker_amd.errors import ConflictError
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    img = Builder(cfg, eng).build()
+    orch = Orchestrator(cfg, eng)
+    name = "clawker.btest.inuse"
+    orch.run(RunOptions(agent="inuse", name=name, image=img, autostart=True,
+                        firewall=False, cmd=["sleep", "10"]))
+    with pytest.raises(ConflictError):
+        eng.remove_image(img)
+    assert eng.images.exists(img)
+    orch.teardown(name, force=True)
+    eng.remove_image(img)          # fine once unused
+    assert not eng.images.exists(img)
+
+
+def test_prune_keeps_fresh_staging_layers(isolated_env):
+    from clawker_amd.engine.images import ImageStore
+    store = ImageStore()
+    lid, fs = store.new_layer_dir()        # fresh tmp- layer (in-flight build)
+    (fs / "f").write_text("x")
+    removed = store.prune_layers()
+    assert (store.root / "layers" / lid).exists()   # survived the prune
