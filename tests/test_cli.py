@@ -238,3 +238,27 @@ def test_init_override_layer_in_subdir(proj):
         assert len(ProjectRegistry().list_projects()) == 1
     finally:
         os.chdir(proj)
+
+
+@requires_isolation
+def test_cli_create_then_start_attach(proj):
+    """create -> start -a: boot plans run and console streams to stdout."""
+    _invoke(["init", "--yes", "--name", "csa", "--harness", "echo"])
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+
+    def clawker(*args, timeout=60, input=None):
+        return subprocess.run([sys.executable, "-m", "clawker_amd", *args],
+                              capture_output=True, text=True, timeout=timeout,
+                              cwd=str(proj), env=env, input=input)
+
+    r = clawker("create", "--agent", "cs", "--no-firewall", "--no-host-services",
+                "--", "/bin/sh", "-c", "echo STARTED-VIA-ATTACH; exit 0")
+    assert r.returncode == 0, r.stderr
+    assert "clawker.csa.cs" in r.stdout
+    r = clawker("ps", "-a", "--format", "json")
+    assert any(x["name"] == "clawker.csa.cs" and x["state"] == "created"
+               for x in json.loads(r.stdout))
+    r = clawker("start", "-a", "cs")
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    assert "STARTED-VIA-ATTACH" in r.stdout
+    clawker("rm", "-f", "cs")
