@@ -34,6 +34,9 @@ def fleet_up(ctx: Ctx, count, branch_prefix, gpus_per_agent, image, base,
     f = ctx.factory
     cfg = f.config(require_project=True)
     fleet = Fleet(cfg, f.orchestrator())
+    if prompt_file:
+        from .prompt import resolve_prompt
+        prompt_file = str(resolve_prompt(prompt_file))
     members = fleet.up(FleetOptions(
         count=count, branch_prefix=branch_prefix, gpus_per_agent=gpus_per_agent,
         cmd=list(cmd), image=image, firewall=firewall, base=base,

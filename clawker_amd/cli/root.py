@@ -78,7 +78,7 @@ def expand_user_alias(argv: list[str]) -> list[str]:
 
 
 def main() -> int:
-    from . import container, doctor, firewall, fleet, image, monitor, project, settings, system, cp, volume, worktree  # noqa
+    from . import container, doctor, firewall, fleet, image, monitor, project, prompt, settings, system, cp, volume, worktree  # noqa
     try:
         cli(args=expand_user_alias(sys.argv[1:]), standalone_mode=False)
         return 0
@@ -102,7 +102,7 @@ def main() -> int:
 
 
 # import groups at module load so `clawker --help` lists them
-from . import container, doctor, firewall, fleet, image, monitor, project, settings, system, cp, volume, worktree  # noqa: E402,F401
+from . import container, doctor, firewall, fleet, image, monitor, project, prompt, settings, system, cp, volume, worktree  # noqa: E402,F401
 
 if __name__ == "__main__":
     # re-enter through the canonical module path so command registration

@@ -262,3 +262,33 @@ def test_cli_create_then_start_attach(proj):
     assert r.returncode == 0, (r.stdout, r.stderr)
     assert "STARTED-VIA-ATTACH" in r.stdout
     clawker("rm", "-f", "cs")
+
+
+def test_prompt_and_bundle_groups(isolated_env, tmp_path):
+    # prompts
+    src = tmp_path / "p.md"
+    src.write_text("do the thing\n")
+    assert _invoke(["prompt", "add", "mytask", str(src)]).exit_code == 0
+    assert "mytask" in _invoke(["prompt", "list"]).output
+    assert "do the thing" in _invoke(["prompt", "show", "mytask"]).output
+    from clawker_amd.cli.prompt import resolve_prompt
+    assert resolve_prompt("mytask").read_text() == "do the thing\n"
+    assert _invoke(["prompt", "rm", "mytask"]).exit_code == 0
+    assert _invoke(["prompt", "show", "mytask"]).exit_code != 0
+    # bundle install (user tier overrides embedded floor)
+    bdir = tmp_path / "myharness"
+    bdir.mkdir()
+    (bdir / "harness.yaml").write_text(
+        "name: myharness\ncmd: [mytool]\nuser: ''\n")
+    (bdir / "evil-link").symlink_to("/etc/passwd")
+    r = _invoke(["bundle", "install", str(bdir)])
+    assert r.exit_code == 0, r.output
+    from clawker_amd import consts
+    installed = consts.config_dir() / "harnesses" / "myharness"
+    assert (installed / "harness.yaml").is_file()
+    assert not (installed / "evil-link").exists()   # symlink sanitized
+    from clawker_amd.bundle import load_harness
+    h = load_harness("myharness")
+    assert h.cmd == ["mytool"]
+    out = _invoke(["bundle", "list"]).output
+    assert "myharness" in out and "user" in out
