@@ -138,3 +138,14 @@ def test_prune_keeps_fresh_staging_layers(isolated_env):
     (fs / "f").write_text("x")
     removed = store.prune_layers()
     assert (store.root / "layers" / lid).exists()   # survived the prune
+
+
+@requires_isolation
+def test_build_layers_carry_no_build_identity(ctx):
+    cfg, eng = ctx
+    from clawker_amd.engine.build import build_layer
+    lid = build_layer(eng, "hostfs", "echo content > /marker")
+    fs = eng.images.layer_path(lid)
+    assert (fs / "marker").exists()
+    assert not (fs / "etc" / "hostname").exists()
+    assert not (fs / "etc" / "hosts").exists()
