@@ -657,3 +657,38 @@ def update_cmd(ctx: Ctx, mem, pids_limit, name):
     if not changed:
         raise ClawkerError("nothing to update (pass --memory/--pids-limit)")
     f.io.success(f"updated {sb}: {', '.join(changed)}")
+
+
+@container_group.command("diff")
+@click.argument("name")
+@pass_factory
+def diff_cmd(ctx: Ctx, name):
+    """Changed files vs the image (the sandbox's copy-on-write upper)."""
+    from pathlib import Path as _P
+    f = ctx.factory
+    sb = resolve_sandbox_name(f, name)
+    info = f.engine().inspect(sb)
+    upper = info.statedir / "upper"
+    if not upper.is_dir():
+        raise ClawkerError(f"no writable layer for {sb} (proc backend?)")
+    skip_prefixes = ("etc/hostname", "etc/hosts", "etc/resolv.conf",
+                     "var/lib/clawker", "run/", "tmp/")
+    rows = []
+    for p in sorted(upper.rglob("*")):
+        rel = str(p.relative_to(upper))
+        if any(rel.startswith(s) for s in skip_prefixes):
+            continue
+        # overlay whiteouts: 0:0 char devices mark deletions
+        try:
+            st = p.lstat()
+        except OSError:
+            continue
+        import stat as _stat
+        if _stat.S_ISCHR(st.st_mode) and st.st_rdev == 0:
+            rows.append(("D", rel))
+        elif p.is_dir():
+            continue
+        else:
+            rows.append(("C", rel))
+    for mark, rel in rows:
+        f.io.print(f"{mark} /{rel}")

@@ -332,3 +332,25 @@ def test_cli_misc_verbs_smoke(proj):
     assert json.loads(r.output)["running"] in (True, False)
     r = _invoke(["system", "prune"])
     assert r.exit_code == 0
+
+
+@requires_isolation
+def test_container_diff(proj):
+    _invoke(["init", "--yes", "--name", "dft", "--harness", "echo"])
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+
+    def clawker(*args, timeout=60):
+        return subprocess.run([sys.executable, "-m", "clawker_amd", *args],
+                              capture_output=True, text=True, timeout=timeout,
+                              cwd=str(proj), env=env)
+
+    r = clawker("run", "-d", "--agent", "df", "--no-firewall",
+                "--no-host-services", "--", "/bin/sh", "-c",
+                "echo x > /newfile; rm /etc/issue 2>/dev/null; sleep 30")
+    assert r.returncode == 0, r.stderr
+    time.sleep(0.5)
+    r = clawker("container", "diff", "df")
+    assert "C /newfile" in r.stdout, r.stdout
+    if "/etc/issue" in r.stdout:
+        assert "D /etc/issue" in r.stdout
+    clawker("rm", "-f", "df")
