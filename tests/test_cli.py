@@ -103,7 +103,7 @@ def test_cli_run_ps_logs_rm_e2e(proj):
     _invoke(["init", "--yes", "--name", "e2e", "--harness", "echo"])
     env = dict(os.environ, PYTHONPATH=str(REPO))
 
-    def clawker(*args, timeout=60):
+    def clawker(*args, timeout=180):
         return subprocess.run([sys.executable, "-m", "clawker_amd", *args],
                               capture_output=True, text=True, timeout=timeout,
                               cwd=str(proj), env=env)
@@ -138,7 +138,7 @@ def test_cli_run_exit_code_propagates(proj):
     r = subprocess.run(
         [sys.executable, "-m", "clawker_amd", "run", "--rm", "--no-firewall",
          "--", "/bin/sh", "-c", "exit 9"],
-        capture_output=True, text=True, timeout=60, cwd=str(proj), env=env)
+        capture_output=True, text=True, timeout=180, cwd=str(proj), env=env)
     assert r.returncode == 9, (r.stdout, r.stderr)
 
 
@@ -165,7 +165,7 @@ def test_cli_cp_copy_roundtrip(proj):
     _invoke(["init", "--yes", "--name", "cptest", "--harness", "echo"])
     env = dict(os.environ, PYTHONPATH=str(REPO))
 
-    def clawker(*args, timeout=60):
+    def clawker(*args, timeout=180):
         return subprocess.run([sys.executable, "-m", "clawker_amd", *args],
                               capture_output=True, text=True, timeout=timeout,
                               cwd=str(proj), env=env)
@@ -246,7 +246,7 @@ def test_cli_create_then_start_attach(proj):
     _invoke(["init", "--yes", "--name", "csa", "--harness", "echo"])
     env = dict(os.environ, PYTHONPATH=str(REPO))
 
-    def clawker(*args, timeout=60, input=None):
+    def clawker(*args, timeout=180, input=None):
         return subprocess.run([sys.executable, "-m", "clawker_amd", *args],
                               capture_output=True, text=True, timeout=timeout,
                               cwd=str(proj), env=env, input=input)
@@ -339,7 +339,7 @@ def test_container_diff(proj):
     _invoke(["init", "--yes", "--name", "dft", "--harness", "echo"])
     env = dict(os.environ, PYTHONPATH=str(REPO))
 
-    def clawker(*args, timeout=60):
+    def clawker(*args, timeout=180):
         return subprocess.run([sys.executable, "-m", "clawker_amd", *args],
                               capture_output=True, text=True, timeout=timeout,
                               cwd=str(proj), env=env)

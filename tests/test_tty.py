@@ -104,7 +104,7 @@ def test_interactive_exec_tty(proj):
     up = subprocess.run(
         [sys.executable, "-m", "clawker_amd", "run", "-d", "--agent", "xt",
          "--no-firewall", "--no-host-services", "--", "sleep", "60"],
-        capture_output=True, text=True, timeout=60, cwd=str(proj), env=env)
+        capture_output=True, text=True, timeout=180, cwd=str(proj), env=env)
     assert up.returncode == 0, up.stderr
     p = subprocess.Popen(
         [sys.executable, "-m", "clawker_amd", "exec", "-i", "-t", "xt", "--",
