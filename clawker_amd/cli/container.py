@@ -48,6 +48,11 @@ def _run_opts_flags(fn):
                       help="skip hostproxy + ssh/gpg agent bridges")(fn)
     fn = click.option("--restart", "restart_policy", default="no",
                       metavar="no|on-failure[:N]", show_default=True)(fn)
+    fn = click.option("-v", "--volume", "volumes_kv", multiple=True,
+                      metavar="SRC:DST[:ro]",
+                      help="bind mount a host path (or named volume) into the sandbox")(fn)
+    fn = click.option("--name", "name_override", default="",
+                      help="full sandbox name (default: clawker.<project>.<agent>)")(fn)
     return fn
 
 
@@ -70,10 +75,35 @@ def _parse_mem(s: str) -> int:
     return int(s)
 
 
+def _parse_volumes(f: Factory, specs) -> list:
+    """-v SRC:DST[:ro] — SRC is a host path (absolute/relative) or a
+    named volume (created on demand)."""
+    from pathlib import Path as _P
+    from ..engine.spec import Mount
+    out = []
+    for sp in specs:
+        parts = sp.split(":")
+        if len(parts) < 2:
+            raise ClawkerError(f"-v expects SRC:DST[:ro], got: {sp}")
+        src, dst = parts[0], parts[1]
+        ro = len(parts) > 2 and parts[2] == "ro"
+        if not dst.startswith("/"):
+            raise ClawkerError(f"-v destination must be absolute: {dst}")
+        if src.startswith("/") or src.startswith(".") or src.startswith("~"):
+            src_path = _P(src).expanduser().resolve()
+            if not src_path.exists():
+                raise ClawkerError(f"-v source missing: {src_path}")
+            out.append(Mount(src=str(src_path), dst=dst, ro=ro))
+        else:
+            vol_path, _ = f.engine().ensure_volume(f"clawker.user.{src}", {})
+            out.append(Mount(src=str(vol_path), dst=dst, ro=ro))
+    return out
+
+
 def _build_opts(f: Factory, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
                 workdir, user, workspace_mode, worktree, firewall, mem,
                 pids_limit, labels_kv, cmd, tty, autostart,
-                no_host_services=False) -> RunOptions:
+                no_host_services=False, volumes_kv=(), name_override="") -> RunOptions:
     cfg = f.config()
     if image == "@":
         image = cfg.image_name()
@@ -84,7 +114,8 @@ def _build_opts(f: Factory, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
         workspace_mode=workspace_mode, firewall=firewall,
         mem_bytes=_parse_mem(mem), pids_max=pids_limit,
         labels=_parse_kv(labels_kv), autostart=autostart,
-        host_services=not no_host_services)
+        host_services=not no_host_services,
+        mounts=_parse_volumes(f, volumes_kv), name=name_override)
     if worktree:
         from ..project.worktrees import ensure_worktree
         wt = ensure_worktree(f.config(require_project=True), worktree)
@@ -137,7 +168,7 @@ def _boot_and_wait(f: Factory, name: str, interactive: bool, tty: bool,
 @pass_factory
 def run_cmd(ctx: Ctx, agent, image, gpus, gpu_indices, hbm_gb, env_kv, workdir,
             user, workspace_mode, worktree, firewall, mem, pids_limit,
-            labels_kv, no_host_services, restart_policy, interactive, tty, detach, rm_after, cmd):
+            labels_kv, no_host_services, restart_policy, volumes_kv, name_override, interactive, tty, detach, rm_after, cmd):
     """Create and start an agent sandbox (alias of `container run`).
 
     CMD may start with an image reference: `clawker run @ -- <cmd>` runs the
@@ -153,7 +184,8 @@ def run_cmd(ctx: Ctx, agent, image, gpus, gpu_indices, hbm_gb, env_kv, workdir,
     opts = _build_opts(f, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
                        workdir, user, workspace_mode, worktree, firewall, mem,
                        pids_limit, labels_kv, cmd, tty, autostart=False,
-                       no_host_services=no_host_services)
+                       no_host_services=no_host_services,
+                       volumes_kv=volumes_kv, name_override=name_override)
     opts.restart = restart_policy
     orch = f.orchestrator()
     info = orch.create(opts)
@@ -172,13 +204,14 @@ container_group.add_command(run_cmd, "run")
 @pass_factory
 def create_cmd(ctx: Ctx, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
                workdir, user, workspace_mode, worktree, firewall, mem,
-               pids_limit, labels_kv, no_host_services, restart_policy, tty, cmd):
+               pids_limit, labels_kv, no_host_services, restart_policy, volumes_kv, name_override, tty, cmd):
     """Create a sandbox without starting it."""
     f = ctx.factory
     opts = _build_opts(f, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
                        workdir, user, workspace_mode, worktree, firewall, mem,
                        pids_limit, labels_kv, list(cmd), tty, autostart=False,
-                       no_host_services=no_host_services)
+                       no_host_services=no_host_services,
+                       volumes_kv=volumes_kv, name_override=name_override)
     opts.restart = restart_policy
     info = f.orchestrator().create(opts)
     f.io.print(info.name)

@@ -354,3 +354,28 @@ def test_container_diff(proj):
     if "/etc/issue" in r.stdout:
         assert "D /etc/issue" in r.stdout
     clawker("rm", "-f", "df")
+
+
+@requires_isolation
+def test_run_with_volume_flag(proj):
+    _invoke(["init", "--yes", "--name", "vfl", "--harness", "echo"])
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    host_dir = proj / "shared-data"
+    host_dir.mkdir()
+    (host_dir / "in.txt").write_text("vol-payload")
+    r = subprocess.run(
+        [sys.executable, "-m", "clawker_amd", "run", "--rm", "--no-firewall",
+         "--no-host-services", "-v", f"{host_dir}:/data",
+         "-v", "scratch:/scratch", "--name", "clawker.vfl.custom", "--",
+         "/bin/sh", "-c",
+         "cat /data/in.txt; echo persisted > /scratch/out.txt"],
+        capture_output=True, text=True, timeout=180, cwd=str(proj), env=env)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    assert "vol-payload" in r.stdout
+    # named volume persisted on the host
+    from clawker_amd.engine import Engine
+    eng = Engine()
+    row = eng.db.get_volume("clawker.user.scratch")
+    assert row is not None
+    assert (Path(row["path"]) / "out.txt").read_text().strip() == "persisted"
+    eng.close()
