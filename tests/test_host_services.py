@@ -52,7 +52,7 @@ def test_open_url_policy_fail_closed(hostproxy, isolated_env):
     assert status == 400
 
 
-def test_callback_register_and_hit(hostproxy):
+def test_callback_register_hit_and_poll(hostproxy):
     status, body = hostproxy.request(
         "POST", "/callback/register",
         json.dumps({"port": 8765, "sandbox": "clawker.t.a"}).encode())
@@ -60,9 +60,19 @@ def test_callback_register_and_hit(hostproxy):
     sess = json.loads(body)
     sid = sess["session"]
     assert f"/cb/{sid}/" in sess["callback_url"]
+    # nothing captured yet
+    status, body = hostproxy.request("GET", f"/callback/poll/{sid}")
+    assert status == 200 and json.loads(body)["hits"] == []
+    # the user's browser hits the callback
     status, body = hostproxy.request("GET", f"/cb/{sid}/done?code=xyz")
     assert status == 200 and b"authentication complete" in body
+    # the agent polls and retrieves the redirect (incl. the auth code)
+    status, body = hostproxy.request("GET", f"/callback/poll/{sid}")
+    hits = json.loads(body)["hits"]
+    assert len(hits) == 1 and hits[0]["path"] == "/done?code=xyz"
     status, _ = hostproxy.request("GET", "/cb/bogus/done")
+    assert status == 404
+    status, _ = hostproxy.request("GET", "/callback/poll/bogus")
     assert status == 404
 
 
