@@ -49,6 +49,44 @@ def fleet_up(ctx: Ctx, count, branch_prefix, gpus_per_agent, image, base,
         _watch(f)
 
 
+@fleet_group.command("run")
+@click.option("-n", "--count", type=int, default=2, show_default=True)
+@click.option("--branch-prefix", default="agent", show_default=True)
+@click.option("--gpus-per-agent", type=int, default=None)
+@click.option("--prompt", "prompt_file", default="")
+@click.option("--no-worktrees", is_flag=True)
+@click.option("--firewall/--no-firewall", "firewall", default=None)
+@click.option("--timeout", type=float, default=3600.0, show_default=True)
+@click.option("--keep", is_flag=True, help="keep sandboxes after completion")
+@click.argument("cmd", nargs=-1, type=click.UNPROCESSED)
+@pass_factory
+def fleet_run(ctx: Ctx, count, branch_prefix, gpus_per_agent, prompt_file,
+              no_worktrees, firewall, timeout, keep, cmd):
+    """up + wait + report (+ down): the one-shot autonomous-loop driver."""
+    f = ctx.factory
+    cfg = f.config(require_project=True)
+    fleet = Fleet(cfg, f.orchestrator())
+    if prompt_file:
+        from .prompt import resolve_prompt
+        prompt_file = str(resolve_prompt(prompt_file))
+    members = fleet.up(FleetOptions(
+        count=count, branch_prefix=branch_prefix, gpus_per_agent=gpus_per_agent,
+        cmd=list(cmd), firewall=firewall, prompt_file=prompt_file,
+        use_worktrees=False if no_worktrees else None))
+    fleet.wait(members, timeout_s=timeout)
+    worst = 0
+    for m in members:
+        mark = "[green]✓[/green]" if m.exit_code == 0 else "[red]✗[/red]"
+        f.io.print(f" {mark} {m.sandbox}  exit={m.exit_code}  "
+                   f"branch={m.branch or '-'}")
+        worst = max(worst, m.exit_code or 0)
+    if not keep:
+        fleet.down(branch_prefix)
+    if worst:
+        from ..errors import ExitError
+        raise ExitError(worst)
+
+
 @fleet_group.command("status")
 @click.option("--branch-prefix", default="agent", show_default=True)
 @click.option("--format", "fmt", default="")

@@ -185,6 +185,21 @@ class CPDaemon:
             (info.name, info.project, info.agent, now, now, info.state))
         self._registry_db.commit()
 
+    def _reconcile_registry(self, infos) -> None:
+        """Keep every known sandbox's row current and close out rows whose
+        sandbox no longer exists in the engine (so `controlplane agents`
+        never shows a removed sandbox as running forever)."""
+        for i in infos:
+            self._record_agent(i)
+        names = {i.name for i in infos}
+        self._registry_db.execute(
+            "UPDATE agents SET state='removed' WHERE state!='removed' "
+            "AND sandbox NOT IN (%s)" % ",".join("?" * len(names))
+            if names else
+            "UPDATE agents SET state='removed' WHERE state!='removed'",
+            tuple(names))
+        self._registry_db.commit()
+
     # ------------------------------------------------------------- watcher --
     def _watch_loop(self) -> None:
         """Track sandbox lifecycle: agent registry, firewall-gateway and
@@ -195,8 +210,7 @@ class CPDaemon:
             try:
                 infos = self.engine.list()
                 running = [i for i in infos if i.state == "running"]
-                for i in running:
-                    self._record_agent(i)
+                self._reconcile_registry(infos)
                 # reconcile firewall gateways against live state (reference:
                 # dockerevents reconcile + FirewallEnable drift guard)
                 live_fw = {i.name: i for i in running

@@ -331,3 +331,27 @@ print("TCPDNS_ANSWERS", ancount)
     assert code == 0
     assert b"P=http://127.0.0.1:3128" in out and b"F=1" in out
     orch.teardown(name, force=True)
+
+
+def test_cp_registry_closes_removed_rows(isolated_env):
+    """The watcher's registry reconcile marks rows for vanished sandboxes
+    'removed' instead of leaving them 'running' forever."""
+    from types import SimpleNamespace
+    from clawker_amd.controlplane.daemon import CPDaemon
+    d = object.__new__(CPDaemon)            # registry-only slice, no daemons
+    d._registry_db = d._open_registry()
+    a = SimpleNamespace(name="clawker.p.a", project="p", agent="a", state="running")
+    b = SimpleNamespace(name="clawker.p.b", project="p", agent="b", state="exited")
+    d._reconcile_registry([a, b])
+    rows = dict(d._registry_db.execute("SELECT sandbox,state FROM agents"))
+    assert rows == {"clawker.p.a": "running", "clawker.p.b": "exited"}
+    # b is removed from the engine entirely; a exits
+    a.state = "exited"
+    d._reconcile_registry([a])
+    rows = dict(d._registry_db.execute("SELECT sandbox,state FROM agents"))
+    assert rows == {"clawker.p.a": "exited", "clawker.p.b": "removed"}
+    # empty engine: everything closes out
+    d._reconcile_registry([])
+    rows = dict(d._registry_db.execute("SELECT sandbox,state FROM agents"))
+    assert set(rows.values()) == {"removed"}
+    d._registry_db.close()

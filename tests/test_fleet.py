@@ -127,3 +127,33 @@ def test_fleet_prompt_file(isolated_env, tmp_path):
         assert "PROMPT:" in logs and "refactor the widget" in logs
     fleet.down()
     fleet.orch.close()
+
+
+def test_fleet_run_one_shot(isolated_env, tmp_path, monkeypatch):
+    """`clawker fleet run` = up + wait + report + down in one verb."""
+    root = tmp_path / "oneshot"
+    root.mkdir()
+    (root / ".clawker.yaml").write_text("project: oneshot\nagent:\n  harness: echo\n")
+    monkeypatch.chdir(root)
+    from click.testing import CliRunner
+    from clawker_amd.cli.root import cli
+    r = CliRunner().invoke(cli, [
+        "fleet", "run", "-n", "2", "--no-worktrees", "--no-firewall",
+        "--timeout", "60", "--", "/bin/sh", "-c", "echo ran-$CLAWKER_AGENT"])
+    if r.exception is not None and not isinstance(r.exception, SystemExit):
+        raise r.exception
+    assert r.exit_code == 0, r.output
+    assert "exit=0" in r.output
+    # sandboxes were torn down (no --keep)
+    from clawker_amd.engine import Engine
+    eng = Engine()
+    assert [i for i in eng.list() if i.project == "oneshot"] == []
+    eng.close()
+    # failing member propagates worst exit code
+    r = CliRunner().invoke(cli, [
+        "fleet", "run", "-n", "1", "--no-worktrees", "--no-firewall",
+        "--timeout", "60", "--", "/bin/sh", "-c", "exit 3"])
+    from clawker_amd.errors import ClawkerError
+    if isinstance(r.exception, ClawkerError):
+        r.exit_code = r.exception.exit_code
+    assert r.exit_code == 3, r.output
