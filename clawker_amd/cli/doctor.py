@@ -84,11 +84,20 @@ def run_checks() -> list[dict]:
 
 @cli.command("doctor")
 @click.option("--format", "fmt", default="")
+@click.option("--collect", "collect_dir", is_flag=False, flag_value=".",
+              default=None,
+              help="write an SOS diagnostic tarball (checks, sandbox "
+                   "forensics, CP events, GPU ledger) to DIR (default .)")
 @pass_factory
-def doctor_cmd(ctx: Ctx, fmt):
+def doctor_cmd(ctx: Ctx, fmt, collect_dir):
     """Diagnose host capabilities for sandboxing + GPU pinning."""
     import json
     f = ctx.factory
+    if collect_dir is not None:
+        from ..sos import collect_bundle
+        out = collect_bundle(Path(collect_dir))
+        f.io.print(str(out))
+        return
     checks = run_checks()
     if fmt == "json":
         f.io.print(json.dumps(checks, indent=1))

@@ -379,3 +379,35 @@ def test_run_with_volume_flag(proj):
     assert row is not None
     assert (Path(row["path"]) / "out.txt").read_text().strip() == "persisted"
     eng.close()
+
+
+def test_doctor_collect_sos_bundle(isolated_env, tmp_path, monkeypatch):
+    """`doctor --collect` ships a forensic tarball with per-sandbox state."""
+    import tarfile
+    ws = tmp_path / "sosproj"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text("project: sostest\n")
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    orch = Orchestrator(load_config(ws))
+    name = "clawker.sostest.a"
+    try:
+        orch.run(RunOptions(agent="a", name=name, autostart=True, firewall=False,
+                            cmd=["/bin/sh", "-c", "echo sos-console-line"]))
+        orch.engine.wait(name, timeout_s=30)
+        r = _invoke(["doctor", "--collect", str(tmp_path / "bundles")])
+        assert r.exit_code == 0, r.output
+        out = Path(r.output.strip().splitlines()[-1])
+        assert out.is_file() and out.name.startswith("clawker-sos-")
+        with tarfile.open(out) as tar:
+            names = tar.getnames()
+            assert "doctor.json" in names and "meta.json" in names
+            assert "sandboxes.json" in names
+            assert f"sandboxes/{name}/status.json" in names
+            console = tar.extractfile(f"sandboxes/{name}/console.log").read()
+            assert b"sos-console-line" in console
+            # no credential material travels
+            assert not [n for n in names if "auth" in n or n.endswith(".key")]
+    finally:
+        orch.teardown(name, force=True)
+        orch.close()
