@@ -80,6 +80,33 @@ def image_rm(ctx: Ctx, force, names):
 cli.add_command(image_rm, "rmi")   # docker-style top-level alias
 
 
+@image_group.command("save")
+@click.option("-o", "--output", default="", help="output tarball (default <name>.tar.gz)")
+@click.argument("name")
+@pass_factory
+def image_save(ctx: Ctx, output, name):
+    """Export an image (manifest + layers) as a tarball (docker save
+    analog — survives the tmpfs layer store across reboots)."""
+    from pathlib import Path
+    f = ctx.factory
+    out = Path(output) if output else Path(
+        name.replace("/", "_").replace(":", "_") + ".tar.gz")
+    p = f.engine().images.save(name, out)
+    f.io.print(str(p))
+
+
+@image_group.command("load")
+@click.option("--name", "rename", default="", help="register under a different name")
+@click.argument("tarball", type=click.Path(exists=True))
+@pass_factory
+def image_load(ctx: Ctx, rename, tarball):
+    """Import a tarball written by `image save`; existing layers are
+    deduplicated by content id."""
+    f = ctx.factory
+    meta = f.engine().images.load(tarball, rename=rename)
+    f.io.print(f"{meta.name}  layers={len(meta.layers)}")
+
+
 @image_group.command("prune")
 @pass_factory
 def image_prune(ctx: Ctx):

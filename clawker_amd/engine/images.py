@@ -17,9 +17,11 @@ Store layout:
 from __future__ import annotations
 
 import hashlib
+import io
 import json
 import os
 import shutil
+import tarfile
 import time
 import uuid
 from dataclasses import dataclass, field
@@ -207,3 +209,56 @@ class ImageStore:
             shutil.rmtree(d, ignore_errors=True)
             removed += 1
         return removed
+
+    # -------------------------------------------------------- save / load --
+    def save(self, name: str, out_path: Path) -> Path:
+        """Export an image (manifest + its layer trees) as a tarball — the
+        docker-save analog. Useful because the layer store defaults to a
+        tmpfs superblock (node reboot loses built images) and for moving
+        images between nodes without rebuilding."""
+        meta = self.get(name)
+        out_path = Path(out_path)
+        out_path.parent.mkdir(parents=True, exist_ok=True)
+        with tarfile.open(out_path, "w:gz") as tar:
+            manifest = json.dumps(meta.to_dict(), indent=1).encode()
+            info = tarfile.TarInfo("manifest.json")
+            info.size = len(manifest)
+            info.mtime = int(time.time())
+            tar.addfile(info, io.BytesIO(manifest))
+            for lid in meta.layers:
+                tar.add(self.root / "layers" / lid, arcname=f"layers/{lid}")
+        return out_path
+
+    def load(self, in_path: Path, rename: str = "") -> ImageMeta:
+        """Import a tarball written by save(). Layers already present
+        (same content id) are skipped; the image record is registered
+        under its saved name (or `rename`)."""
+        self.ensure_layer_filesystem()
+        with tarfile.open(in_path) as tar:
+            names = tar.getnames()
+            if "manifest.json" not in names:
+                raise ConflictError(f"{in_path}: not a clawker image tarball")
+            # refuse traversal: every member stays under layers/ or is the
+            # manifest (absolute paths and .. segments rejected)
+            for n in names:
+                parts = Path(n).parts
+                if (n != "manifest.json" and parts[:1] != ("layers",)) \
+                        or ".." in parts or n.startswith("/"):
+                    raise ConflictError(f"{in_path}: unsafe member {n!r}")
+            meta = ImageMeta.from_dict(
+                json.loads(tar.extractfile("manifest.json").read()))
+            for lid in meta.layers:
+                dst = self.root / "layers" / lid
+                if dst.exists():
+                    continue
+                members = [m for m in tar.getmembers()
+                           if Path(m.name).parts[:2] == ("layers", lid)]
+                if not members:
+                    raise ConflictError(
+                        f"{in_path}: layer {lid} missing from tarball")
+                tar.extractall(self.root, members=members, numeric_owner=True)
+        if rename:
+            meta.name = rename
+        meta.created = time.time()
+        self.put(meta)
+        return meta

@@ -149,3 +149,48 @@ def test_build_layers_carry_no_build_identity(ctx):
     assert (fs / "marker").exists()
     assert not (fs / "etc" / "hostname").exists()
     assert not (fs / "etc" / "hosts").exists()
+
+
+@requires_isolation
+def test_image_save_load_roundtrip(ctx, tmp_path):
+    """save -> wipe -> load restores the image byte-for-byte (layer ids
+    are content-addressed, so dedup on load is safe)."""
+    import tarfile
+    cfg, eng = ctx
+    from clawker_amd.engine.build import build_image
+    build_image(eng, "saveme:latest", "hostfs",
+                "echo payload > /saved-marker")
+    name = "saveme:latest"
+    meta = eng.images.get(name)
+    assert meta.layers
+    out = eng.images.save(name, tmp_path / "img.tar.gz")
+    with tarfile.open(out) as tar:
+        assert "manifest.json" in tar.getnames()
+    # wipe the image + its layers, then load
+    eng.images.remove(name)
+    assert not eng.images.exists(name)
+    loaded = eng.images.load(out)
+    assert loaded.name == name
+    assert loaded.layers == meta.layers
+    for lid in loaded.layers:
+        assert eng.images.layer_path(lid).is_dir()
+    # load again (layers present): pure no-op dedup, still registered
+    eng.images.load(out, rename="copy:latest")
+    assert eng.images.get("copy:latest").layers == meta.layers
+
+
+def test_image_load_rejects_traversal(isolated_env, tmp_path):
+    import io
+    import tarfile
+    import pytest as _pytest
+    from clawker_amd.engine.images import ImageStore
+    from clawker_amd.errors import ConflictError
+    evil = tmp_path / "evil.tar.gz"
+    with tarfile.open(evil, "w:gz") as tar:
+        m = json.dumps({"name": "evil:latest", "layers": ["x"]}).encode()
+        ti = tarfile.TarInfo("manifest.json"); ti.size = len(m)
+        tar.addfile(ti, io.BytesIO(m))
+        ti = tarfile.TarInfo("layers/../../etc/cron.d/pwn"); ti.size = 0
+        tar.addfile(ti, io.BytesIO(b""))
+    with _pytest.raises(ConflictError, match="unsafe"):
+        ImageStore().load(evil)
