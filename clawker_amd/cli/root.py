@@ -79,6 +79,13 @@ def expand_user_alias(argv: list[str]) -> list[str]:
 
 def main() -> int:
     from . import container, doctor, firewall, fleet, image, monitor, project, prompt, settings, system, cp, volume, worktree  # noqa
+    try:       # once-per-version changelog teaser (interactive runs only)
+        if sys.stderr.isatty():
+            from ..iostreams import IOStreams
+            from ..update import maybe_show_teaser
+            maybe_show_teaser(IOStreams())
+    except Exception:
+        pass
     try:
         cli(args=expand_user_alias(sys.argv[1:]), standalone_mode=False)
         return 0

@@ -411,3 +411,24 @@ def test_doctor_collect_sos_bundle(isolated_env, tmp_path, monkeypatch):
     finally:
         orch.teardown(name, force=True)
         orch.close()
+
+
+def test_changelog_teaser_once_per_version(isolated_env):
+    """Version bump -> one teaser from the local CHANGELOG, cursor advances
+    (reference: update-notifier + changelog teaser, air-gap redesign)."""
+    from clawker_amd import update
+    from clawker_amd.iostreams import TestIOStreams
+    io = TestIOStreams()
+    # first run ever: quiet, cursor written
+    assert update.maybe_show_teaser(io) is False
+    assert update.maybe_show_teaser(io) is False      # same version: quiet
+    # simulate an upgrade: rewind the cursor to an older version
+    update._cursor_path().write_text('{"last_version": "0.0.9"}')
+    assert update.maybe_show_teaser(io) is True
+    err = io.err
+    assert "0.0.9 → " in err and "new in this version" in err
+    assert update.maybe_show_teaser(io) is False      # shown once
+    # teaser content comes from the current version's section
+    lines = update.teaser_for("0.1.0")
+    assert lines and len(lines) <= 6
+    assert update.teaser_for("99.99.99") == []
