@@ -653,6 +653,31 @@ def restart_cmd(ctx: Ctx, timeout, names):
 container_group.add_command(restart_cmd, "restart")
 
 
+@container_group.command("pause")
+@click.argument("names", nargs=-1, required=True)
+@pass_factory
+def pause_cmd(ctx: Ctx, names):
+    """Freeze a sandbox's workload (SIGSTOP the tree under its init;
+    the supervisor stays live so status/unpause keep working)."""
+    f = ctx.factory
+    for n in names:
+        sb = resolve_sandbox_name(f, n)
+        f.engine().pause(sb)
+        f.io.print(sb)
+
+
+@container_group.command("unpause")
+@click.argument("names", nargs=-1, required=True)
+@pass_factory
+def unpause_cmd(ctx: Ctx, names):
+    """Thaw a paused sandbox (SIGCONT)."""
+    f = ctx.factory
+    for n in names:
+        sb = resolve_sandbox_name(f, n)
+        f.engine().unpause(sb)
+        f.io.print(sb)
+
+
 @container_group.command("update")
 @click.option("--memory", "-m", "mem", default="", help="new memory limit (e.g. 8g)")
 @click.option("--pids-limit", type=int, default=0)

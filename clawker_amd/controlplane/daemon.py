@@ -209,7 +209,7 @@ class CPDaemon:
         while not self._stop.is_set():
             try:
                 infos = self.engine.list()
-                running = [i for i in infos if i.state == "running"]
+                running = [i for i in infos if i.state in ("running", "paused")]
                 self._reconcile_registry(infos)
                 # reconcile firewall gateways against live state (reference:
                 # dockerevents reconcile + FirewallEnable drift guard)

@@ -9,7 +9,7 @@ from pathlib import Path
 import pytest
 
 from clawker_amd import consts
-from clawker_amd.engine import Engine, SandboxSpec
+from clawker_amd.engine import Engine, Mount, SandboxSpec
 from clawker_amd.engine.images import HOSTFS, ImageStore
 from clawker_amd.engine.state import StateDB
 from clawker_amd.errors import ConflictError, NotFoundError
@@ -205,3 +205,37 @@ def test_image_store_layers(isolated_env, tmp_path):
     assert store.layer_path(final)         # still there
     with pytest.raises(Exception):
         store.layer_path(final2)
+
+
+@requires_isolation
+def test_pause_unpause_freezes_workload(engine, tmp_path):
+    """pause SIGSTOPs the tree under ckd (counter stops advancing);
+    unpause resumes it; ps state surfaces 'paused'; stop still works."""
+    import time as _t
+    out = tmp_path / "count"
+    out.touch()
+    info = engine.create(_spec(
+        "pz", f"i=0; while true; do i=$((i+1)); echo $i > {out}; "
+              "sleep 0.05; done",
+        mounts=[Mount(src=str(tmp_path), dst=str(tmp_path), ro=False)]))
+    engine.start(info.name)
+    deadline = _t.monotonic() + 10
+    while out.read_text().strip() in ("", "0") and _t.monotonic() < deadline:
+        _t.sleep(0.02)
+    assert engine.pause(info.name) >= 1
+    assert engine.inspect(info.name).state == "paused"
+    v1 = out.read_text()
+    _t.sleep(0.4)
+    assert out.read_text() == v1          # frozen: no progress
+    engine.unpause(info.name)
+    assert engine.inspect(info.name).state == "running"
+    deadline = _t.monotonic() + 10
+    while out.read_text() == v1 and _t.monotonic() < deadline:
+        _t.sleep(0.02)
+    assert out.read_text() != v1          # thawed: progress resumes
+    engine.stop(info.name, timeout_s=5)
+    assert engine.inspect(info.name).state == "exited"
+    # restart clears the paused marker
+    engine.start(info.name)
+    assert engine.inspect(info.name).state == "running"
+    engine.remove(info.name, force=True)
