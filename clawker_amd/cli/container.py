@@ -653,6 +653,21 @@ def restart_cmd(ctx: Ctx, timeout, names):
 container_group.add_command(restart_cmd, "restart")
 
 
+@container_group.command("commit")
+@click.option("-m", "--message", default="", help="commit annotation label")
+@click.argument("name")
+@click.argument("image")
+@pass_factory
+def commit_cmd(ctx: Ctx, message, name, image):
+    """Snapshot a sandbox's writable layer as a new image (docker commit
+    analog: agent-installed tools become a reusable layer)."""
+    f = ctx.factory
+    sb = resolve_sandbox_name(f, name)
+    from ..engine.build import commit_sandbox
+    meta = commit_sandbox(f.engine(), sb, image, message=message)
+    f.io.print(f"{meta.name}  layers={len(meta.layers)}")
+
+
 @container_group.command("pause")
 @click.argument("names", nargs=-1, required=True)
 @pass_factory

@@ -6,8 +6,11 @@ here the engine is its own builder (no daemon): layer = the COW residue of
 the build sandbox. Requires the ns backend (no overlay on proc hosts)."""
 from __future__ import annotations
 
+import shutil
+import subprocess
 import time
 import uuid
+from pathlib import Path
 from typing import Callable
 
 from .. import consts
@@ -94,4 +97,45 @@ def build_image(engine: Engine, name: str, base_image: str, script: str,
         parent=base_image, base_hash=base_hash)
     engine.images.put(meta)
     log.info("image_built", image=name, layer=lid)
+    return meta
+
+
+def commit_sandbox(engine: Engine, name: str, image_name: str,
+                   message: str = "") -> ImageMeta:
+    """docker-commit analog: snapshot a sandbox's copy-on-write upper as
+    a new content-addressed layer stacked on its image. Overlay artifacts
+    (whiteout device nodes, opaque-dir xattrs) are preserved via cp -a so
+    deletions carry into the committed image; runtime residue (rundir
+    mountpoint, identity files rewritten per-create anyway) is scrubbed."""
+    info = engine.inspect(name)
+    upper = Path(info.statedir) / "upper"
+    if not upper.is_dir():
+        raise EngineError("commit", f"no writable layer for {name}")
+    tmp_id, layer_fs = engine.images.new_layer_dir()
+    r = subprocess.run(
+        ["cp", "-a", "--reflink=auto", f"{upper}/.", str(layer_fs)],
+        capture_output=True, text=True)
+    if r.returncode != 0:
+        shutil.rmtree(layer_fs.parent, ignore_errors=True)
+        raise EngineError("commit", f"layer copy failed: {r.stderr.strip()}")
+    for junk in ("run/clawker", ".oldroot", "etc/hostname", "etc/hosts",
+                 "etc/resolv.conf"):
+        p = layer_fs / junk
+        if p.is_dir() and not p.is_symlink():
+            shutil.rmtree(p, ignore_errors=True)
+        else:
+            p.unlink(missing_ok=True)
+    lid = engine.images.commit_layer(tmp_id)
+    base = engine.images.get(info.image)
+    # layers holds ONLY this image's own layer: lowerdirs_for composes the
+    # full stack by walking the parent chain (duplicating base layers here
+    # would mount the same lowerdir twice -> overlayfs ELOOP)
+    meta = ImageMeta(
+        name=image_name, layers=[lid], env=dict(base.env),
+        user=base.user, cmd=list(base.cmd), workdir=base.workdir,
+        labels={**base.labels, "dev.clawker.commit.from": name,
+                **({"dev.clawker.commit.message": message} if message else {})},
+        parent=info.image)
+    engine.images.put(meta)
+    log.info("sandbox_committed", sandbox=name, image=image_name, layer=lid)
     return meta

@@ -211,22 +211,44 @@ class ImageStore:
         return removed
 
     # -------------------------------------------------------- save / load --
+    def _image_chain(self, name: str) -> list[ImageMeta]:
+        """The image plus its parents up to (not incl.) hostfs."""
+        chain: list[ImageMeta] = []
+        seen: set[str] = set()
+        cur: ImageMeta | None = self.get(name)
+        while cur is not None:
+            chain.append(cur)
+            parent = cur.parent
+            if not parent or parent == HOSTFS or parent in seen:
+                break
+            seen.add(parent)
+            cur = self.get(parent)
+        return chain
+
     def save(self, name: str, out_path: Path) -> Path:
-        """Export an image (manifest + its layer trees) as a tarball — the
-        docker-save analog. Useful because the layer store defaults to a
-        tmpfs superblock (node reboot loses built images) and for moving
-        images between nodes without rebuilding."""
-        meta = self.get(name)
+        """Export an image — manifest for it AND its parent chain, plus
+        every referenced layer tree — as a tarball (docker-save analog).
+        Useful because the layer store defaults to a tmpfs superblock
+        (node reboot loses built images) and for moving images between
+        nodes without rebuilding."""
+        chain = self._image_chain(name)
         out_path = Path(out_path)
         out_path.parent.mkdir(parents=True, exist_ok=True)
         with tarfile.open(out_path, "w:gz") as tar:
-            manifest = json.dumps(meta.to_dict(), indent=1).encode()
+            manifest = json.dumps(
+                {"image": name, "images": [m.to_dict() for m in chain]},
+                indent=1).encode()
             info = tarfile.TarInfo("manifest.json")
             info.size = len(manifest)
             info.mtime = int(time.time())
             tar.addfile(info, io.BytesIO(manifest))
-            for lid in meta.layers:
-                tar.add(self.root / "layers" / lid, arcname=f"layers/{lid}")
+            done: set[str] = set()
+            for m in chain:
+                for lid in m.layers:
+                    if lid not in done:
+                        done.add(lid)
+                        tar.add(self.root / "layers" / lid,
+                                arcname=f"layers/{lid}")
         return out_path
 
     def load(self, in_path: Path, rename: str = "") -> ImageMeta:
@@ -245,20 +267,31 @@ class ImageStore:
                 if (n != "manifest.json" and parts[:1] != ("layers",)) \
                         or ".." in parts or n.startswith("/"):
                     raise ConflictError(f"{in_path}: unsafe member {n!r}")
-            meta = ImageMeta.from_dict(
-                json.loads(tar.extractfile("manifest.json").read()))
-            for lid in meta.layers:
-                dst = self.root / "layers" / lid
-                if dst.exists():
-                    continue
-                members = [m for m in tar.getmembers()
-                           if Path(m.name).parts[:2] == ("layers", lid)]
-                if not members:
-                    raise ConflictError(
-                        f"{in_path}: layer {lid} missing from tarball")
-                tar.extractall(self.root, members=members, numeric_owner=True)
+            manifest = json.loads(tar.extractfile("manifest.json").read())
+            if "images" in manifest:      # chain format
+                metas = [ImageMeta.from_dict(d) for d in manifest["images"]]
+                top = next(m for m in metas if m.name == manifest["image"])
+            else:                         # single-image format
+                metas = [ImageMeta.from_dict(manifest)]
+                top = metas[0]
+            for meta in metas:
+                for lid in meta.layers:
+                    dst = self.root / "layers" / lid
+                    if dst.exists():
+                        continue
+                    members = [m for m in tar.getmembers()
+                               if Path(m.name).parts[:2] == ("layers", lid)]
+                    if not members:
+                        raise ConflictError(
+                            f"{in_path}: layer {lid} missing from tarball")
+                    tar.extractall(self.root, members=members,
+                                   numeric_owner=True)
         if rename:
-            meta.name = rename
-        meta.created = time.time()
-        self.put(meta)
-        return meta
+            top.name = rename
+            # parents keep their names: the renamed top still chains to them
+        for meta in metas[::-1]:          # parents first
+            if meta is not top and self.exists(meta.name):
+                continue                  # never clobber an existing parent
+            meta.created = time.time()
+            self.put(meta)
+        return top

@@ -177,6 +177,15 @@ def test_image_save_load_roundtrip(ctx, tmp_path):
     # load again (layers present): pure no-op dedup, still registered
     eng.images.load(out, rename="copy:latest")
     assert eng.images.get("copy:latest").layers == meta.layers
+    # chained image: save carries the parent chain too
+    from clawker_amd.engine.build import build_image as _bi
+    _bi(eng, "child:latest", name, "echo two > /two-marker")
+    out2 = eng.images.save("child:latest", tmp_path / "chain.tar.gz")
+    eng.images.remove("child:latest")
+    eng.images.remove(name)
+    top = eng.images.load(out2)
+    assert top.name == "child:latest" and eng.images.exists(name)
+    assert len(eng.images.lowerdirs_for("child:latest")) == 3
 
 
 def test_image_load_rejects_traversal(isolated_env, tmp_path):
@@ -194,3 +203,41 @@ def test_image_load_rejects_traversal(isolated_env, tmp_path):
         tar.addfile(ti, io.BytesIO(b""))
     with _pytest.raises(ConflictError, match="unsafe"):
         ImageStore().load(evil)
+
+
+@requires_isolation
+def test_container_commit_snapshot_layer(ctx):
+    """A sandbox's writes (incl. a deletion whiteout) become a reusable
+    image layer; new sandboxes from the committed image see both."""
+    cfg, eng = ctx
+    from clawker_amd import consts
+    from clawker_amd.engine import SandboxSpec
+    from clawker_amd.engine.build import build_image, commit_sandbox
+    build_image(eng, "cbase:latest", "hostfs",
+                "echo original > /victim.txt")
+    name = consts.SANDBOX_NAME_PREFIX + "btest.committer"
+    eng.create(SandboxSpec(
+        name=name, hostname="sbx", autostart=True, netns=True,
+        cmd=["/bin/sh", "-c", "echo tooled > /opt/agent-tool; rm /victim.txt"]),
+        image="cbase:latest")
+    eng.start(name)
+    assert eng.wait(name, timeout_s=15) == 0
+    meta = commit_sandbox(eng, name, "committed:latest", message="tools")
+    assert meta.parent == "cbase:latest" and len(meta.layers) == 1
+    # full stack resolves through the parent chain without duplicates
+    lowers = eng.images.lowerdirs_for("committed:latest")
+    assert len(lowers) == len(set(lowers)) == 3
+    assert meta.labels["dev.clawker.commit.message"] == "tools"
+    # run from the committed image: the write is there, the delete holds
+    name2 = consts.SANDBOX_NAME_PREFIX + "btest.fromcommit"
+    eng.create(SandboxSpec(
+        name=name2, hostname="sbx", autostart=True, netns=True,
+        cmd=["/bin/sh", "-c",
+             "cat /opt/agent-tool; test ! -e /victim.txt && echo gone"]),
+        image="committed:latest")
+    eng.start(name2)
+    assert eng.wait(name2, timeout_s=15) == 0
+    logs = eng.logs(name2).decode()
+    assert "tooled" in logs and "gone" in logs
+    eng.remove(name, force=True)
+    eng.remove(name2, force=True)
