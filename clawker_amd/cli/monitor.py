@@ -9,10 +9,14 @@ Prometheus exporter daemon."""
 from __future__ import annotations
 
 import json
+import shutil
 import time
 
 import click
 
+from .. import consts
+from ..cmdutil import format_age
+from ..errors import ClawkerError
 from .root import Ctx, cli, pass_factory
 
 
@@ -134,6 +138,56 @@ def dashboard_cmd(ctx: Ctx, interval):
                 time.sleep(interval)
         except KeyboardInterrupt:
             pass
+
+
+@monitor_group.command("extensions")
+@click.option("--install", "install_file", default=None,
+              type=click.Path(exists=True, dir_okay=False),
+              help="drop a .prom collector file into the metrics.d lane")
+@click.option("--remove", "remove_name", default="",
+              help="remove an installed collector by name")
+@pass_factory
+def monitor_extensions(ctx: Ctx, install_file, remove_name):
+    """Textfile-collector extensions (reference: monitoring units plugin
+    model — here: .prom files in <state>/metrics.d/ merged into /metrics)."""
+    import time as _t
+    from pathlib import Path as _P
+    f = ctx.factory
+    metrics_d = consts.state_dir() / "metrics.d"
+    if install_file:
+        src = _P(install_file)
+        if src.suffix != ".prom":
+            raise ClawkerError("collector files must end in .prom")
+        metrics_d.mkdir(parents=True, exist_ok=True)
+        shutil.copy2(src, metrics_d / src.name)
+        f.io.success(f"installed {src.name} -> {metrics_d}")
+        return
+    if remove_name:
+        p = metrics_d / (remove_name if remove_name.endswith(".prom")
+                         else remove_name + ".prom")
+        if not p.is_file():
+            raise ClawkerError(f"no such collector: {p.name}")
+        p.unlink()
+        f.io.success(f"removed {p.name}")
+        return
+    files = sorted(metrics_d.glob("*.prom")) if metrics_d.is_dir() else []
+    if not files:
+        f.io.eprint(f"no extensions ({metrics_d}/*.prom)")
+        return
+    from rich.table import Table
+    t = Table(box=None, pad_edge=False)
+    for c in ("NAME", "METRICS", "AGE"):
+        t.add_column(c)
+    for p in files:
+        try:
+            text = p.read_text()
+            n = sum(1 for l in text.splitlines()
+                    if l.strip() and not l.startswith("#"))
+            age = format_age(_t.time() - p.stat().st_mtime)
+        except OSError:
+            n, age = 0, "?"
+        t.add_row(p.name, str(n), age)
+    f.io.print(t)
 
 
 @monitor_group.command("serve", hidden=True)

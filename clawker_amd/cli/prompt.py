@@ -136,3 +136,9 @@ def bundle_list(ctx: Ctx):
             tier = "user"
         t.add_row("harness", name, tier)
     f.io.print(t)
+
+
+# reference parity: the bundle tier is also exposed as `plugin` with a
+# `skill` alias (internal/cmd plugin group, alias skill — SURVEY.md §2.1)
+cli.add_command(bundle_group, "plugin")
+cli.add_command(bundle_group, "skill")

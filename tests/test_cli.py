@@ -432,3 +432,25 @@ def test_changelog_teaser_once_per_version(isolated_env):
     lines = update.teaser_for("0.1.0")
     assert lines and len(lines) <= 6
     assert update.teaser_for("99.99.99") == []
+
+
+def test_monitor_extensions_lane(proj, tmp_path):
+    """metrics.d textfile-collector verbs (monitoring units analog)."""
+    prom = tmp_path / "agent_cost.prom"
+    prom.write_text("# HELP agent_cost_usd total spend\nagent_cost_usd 1.25\n")
+    r = _invoke(["monitor", "extensions", "--install", str(prom)])
+    assert r.exit_code == 0
+    r = _invoke(["monitor", "extensions"])
+    assert "agent_cost.prom" in r.output and "1" in r.output
+    # merged into /metrics by the exporter
+    from clawker_amd.monitor.exporter import _metrics_text
+    assert "agent_cost_usd 1.25" in _metrics_text()
+    r = _invoke(["monitor", "extensions", "--remove", "agent_cost"])
+    assert r.exit_code == 0
+    r = _invoke(["monitor", "extensions", "--remove", "agent_cost"])
+    assert r.exit_code != 0
+    # plugin/skill aliases resolve to the bundle group
+    r = _invoke(["plugin", "list"])
+    assert r.exit_code == 0
+    r = _invoke(["skill", "list"])
+    assert r.exit_code == 0
