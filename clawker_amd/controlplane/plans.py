@@ -44,10 +44,13 @@ def _exists_in_sandbox(client: CkdClient, path: str) -> bool:
 
 
 def run_plan(client: CkdClient, steps: list[Step],
-             on_step: Callable[[str, int], None] | None = None) -> None:
+             on_step: Callable[[str, int], None] | None = None,
+             on_start: Callable[[str], None] | None = None) -> None:
     for step in steps:
         if step.skip_if_missing and not _exists_in_sandbox(client, step.skip_if_missing):
             continue
+        if on_start:
+            on_start(step.name)
         stage: dict = {"argv": step.argv}
         if step.user:
             stage["user"] = step.user
@@ -75,20 +78,26 @@ def boot_plan() -> list[Step]:
 
 
 def drive_boot(client: CkdClient, hello: dict,
-               on_step: Callable[[str, int], None] | None = None) -> None:
+               on_step: Callable[[str, int], None] | None = None,
+               on_start: Callable[[str], None] | None = None) -> None:
     """Drive InitPlan (first boot only) then BootPlan, matching the
     reference's CP dial flow (SURVEY.md §3.1 lower half)."""
     if not hello.get("initialized"):
-        run_plan(client, init_plan(), on_step)
+        run_plan(client, init_plan(), on_step, on_start)
         client.agent_initialized()
-    run_plan(client, boot_plan(), on_step)
+    run_plan(client, boot_plan(), on_step, on_start)
 
 
 def run_boot_plans(factory, name: str, client: CkdClient, hello: dict,
                    quiet: bool = False) -> None:
+    def on_start(step: str) -> None:
+        if not quiet:
+            factory.io.eprint(f"  [dim]»[/dim] {step} ...")
+
     def on_step(step: str, code: int) -> None:
         if not quiet:
-            mark = "✓" if code == 0 else "!"
+            mark = "[green]✓[/green]" if code == 0 else "[yellow]![/yellow]"
             factory.io.eprint(f"  {mark} {step}")
 
-    drive_boot(client, hello, None if quiet else on_step)
+    drive_boot(client, hello, None if quiet else on_step,
+               None if quiet else on_start)
