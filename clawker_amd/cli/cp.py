@@ -63,14 +63,9 @@ def cp_events(ctx: Ctx, n, follow):
         seen = max(seen, int(ev.get("ts", 0) * 1e6))
     if not follow:
         return
-    import time as _t
-    while True:
-        _t.sleep(1.0)
-        for ev in cp.events(200):
-            ts = int(ev.get("ts", 0) * 1e6)
-            if ts > seen:
-                ctx.factory.io.print(json.dumps(ev))
-                seen = max(seen, ts)
+    # live push stream via cpd pub/sub (reference: Topic subscriber)
+    for ev in cp.follow_events():
+        ctx.factory.io.print(json.dumps(ev))
 
 
 @cp_group.command("serve", hidden=True)

@@ -122,3 +122,23 @@ class CPClient:
 
     def events(self, n: int = 100) -> list[dict]:
         return self.request({"op": "events", "n": n})["events"]
+
+    def follow_events(self):
+        """Generator of live events pushed by cpd's pub/sub (no polling).
+        Ends when the daemon goes away; caller breaks to disconnect."""
+        if self.auto_start:
+            self.ensure_running()
+        s = self._connect()
+        try:
+            wire.send_frame(s, {"op": "events_follow"})
+            hello = wire.recv_frame(s)
+            if not (hello and hello.get("ok")):
+                raise ClawkerError(
+                    f"control plane error: {(hello or {}).get('error')}")
+            while True:
+                frame = wire.recv_frame(s)
+                if frame is None:
+                    return
+                yield frame["event"]
+        finally:
+            s.close()

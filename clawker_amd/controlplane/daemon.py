@@ -54,10 +54,11 @@ def events_path() -> Path:
 class EventLog:
     """Append-only jsonl event stream (the pub/sub + OTLP lanes analog)."""
 
-    def __init__(self, path: Path):
+    def __init__(self, path: Path, on_emit=None):
         self.path = path
         self.path.parent.mkdir(parents=True, exist_ok=True)
         self._lock = threading.Lock()
+        self.on_emit = on_emit     # live pub/sub fan-out (Topic.publish)
 
     MAX_BYTES = 10 * 1024 * 1024
 
@@ -71,6 +72,8 @@ class EventLog:
                 pass
             with open(self.path, "a") as f:
                 f.write(json.dumps(rec) + "\n")
+        if self.on_emit is not None:
+            self.on_emit(rec)
 
     def tail(self, n: int = 100) -> list[dict]:
         try:
@@ -145,9 +148,11 @@ class CPDaemon:
         from ..config.config import load_settings
         from ..engine import Engine
         from ..firewall.gateway import GatewayManager
+        from .pubsub import Topic
         self.settings = load_settings().get()
         self.engine = Engine()
-        self.events = EventLog(events_path())
+        self.topic = Topic("cp-events")
+        self.events = EventLog(events_path(), on_emit=self.topic.publish)
         self.queue = ActionQueue()
         from ..socketbridge import SocketBridgeManager
         self.gateways = GatewayManager(
@@ -357,12 +362,32 @@ class CPDaemon:
             threading.Thread(target=self._serve_conn, args=(conn,),
                              daemon=True).start()
 
+    def _follow_events(self, conn: socket.socket, req: dict) -> None:
+        """Stream events as frames until the client goes away (reference:
+        pub/sub subscriber semantics: bounded buffer, drop-oldest)."""
+        sub = self.topic.subscribe()
+        try:
+            wire.send_frame(conn, {"ok": True, "stream": True})
+            while not self._stop.is_set():
+                ev = sub.get(timeout=1.0)
+                if ev is None:
+                    continue
+                wire.send_frame(conn, {"event": ev, "dropped": sub.dropped})
+        except OSError:
+            pass          # client disconnected
+        finally:
+            sub.close()
+
     def _serve_conn(self, conn: socket.socket) -> None:
         try:
             conn.settimeout(30)
             while True:
                 req = wire.recv_frame(conn)
                 if req is None:
+                    return
+                if req.get("op") == "events_follow":
+                    conn.settimeout(None)
+                    self._follow_events(conn, req)
                     return
                 try:
                     resp = self._handle_admin(req)
