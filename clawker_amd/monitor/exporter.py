@@ -7,6 +7,7 @@ samplers — no collector chain.
 """
 from __future__ import annotations
 
+import json
 import os
 import signal
 import subprocess
@@ -24,6 +25,62 @@ log = get_logger("exporter")
 
 def pid_path() -> Path:
     return consts.runtime_dir() / "exporter.pid"
+
+
+class _EventCounters:
+    """Incremental tail of cp-events.jsonl -> monotonic counters (the
+    netlogger/OTLP egress-decision lane, re-landed as Prometheus metrics:
+    this node's observability sink is the scrape endpoint, not a
+    collector). Rotation-aware: the 10MB rollover resets the offset."""
+
+    def __init__(self):
+        self.offset = 0
+        self.ino = None
+        self.events: dict[str, int] = {}
+        self.egress: dict[tuple, int] = {}
+
+    def update(self) -> None:
+        from ..controlplane.daemon import events_path
+        path = events_path()
+        try:
+            st = path.stat()
+        except OSError:
+            return
+        if st.st_ino != self.ino or st.st_size < self.offset:
+            self.ino, self.offset = st.st_ino, 0    # new/rotated file
+        try:
+            with open(path) as f:
+                f.seek(self.offset)
+                chunk = f.read()
+                self.offset = f.tell()
+        except OSError:
+            return
+        for line in chunk.splitlines():
+            try:
+                rec = json.loads(line)
+            except ValueError:
+                continue
+            ev = rec.get("event", "unknown")
+            self.events[ev] = self.events.get(ev, 0) + 1
+            if ev == "egress_decision":
+                key = (rec.get("sandbox", ""), rec.get("action", ""),
+                       rec.get("proto", ""))
+                self.egress[key] = self.egress.get(key, 0) + 1
+
+    def lines(self) -> list[str]:
+        out = ["# TYPE clawker_cp_events_total counter"]
+        for ev, n in sorted(self.events.items()):
+            out.append(f'clawker_cp_events_total{{event="{ev}"}} {n}')
+        if self.egress:
+            out.append("# TYPE clawker_egress_decisions_total counter")
+            for (sb, action, proto), n in sorted(self.egress.items()):
+                out.append(
+                    f'clawker_egress_decisions_total{{sandbox="{sb}",'
+                    f'action="{action}",proto="{proto}"}} {n}')
+        return out
+
+
+_event_counters = _EventCounters()
 
 
 def _metrics_text() -> str:
@@ -62,6 +119,8 @@ def _metrics_text() -> str:
         if s.get("pids"):
             lines.append(f"clawker_sandbox_pids{{{lbl}}} {s['pids']}")
     lines.append(f"clawker_sandboxes_running {running}")
+    _event_counters.update()
+    lines.extend(_event_counters.lines())
     # plugin lane: merge third-party textfile metrics (the reference's
     # monitoring-units collector routing, reduced to the node-exporter
     # textfile pattern: drop .prom files into <state>/metrics.d/)
