@@ -27,6 +27,32 @@ def build_cmd(ctx: Ctx, harness, no_cache, quiet):
     cfg = f.config(require_project=True)
     from ..bundler import Builder
     builder = Builder(cfg, f.engine())
+    if not quiet and f.io.is_stderr_tty():
+        # live progress tree (tui components; reference: RunProgress)
+        from ..tui.components import ProgressSteps
+        from rich.live import Live
+        steps = ProgressSteps()
+        cur: list = [""]
+
+        def progress(line: str) -> None:
+            if line.startswith("building "):
+                if cur[0]:
+                    steps.done(cur[0])
+                cur[0] = line.split(" image ", 1)[-1]
+                steps.start(cur[0])
+            elif cur[0]:
+                steps.start(cur[0], detail=line[-60:])
+            live.refresh()
+
+        with Live(steps, console=f.io.err_console,
+                  refresh_per_second=8) as live:
+            name = builder.build(harness_name=harness, no_cache=no_cache,
+                                 on_progress=progress)
+            if cur[0]:
+                steps.done(cur[0], detail="")
+            live.refresh()
+        f.io.print(name)
+        return
     progress = None if quiet else (lambda line: f.io.eprint(f"[dim]»[/dim] {line}"))
     name = builder.build(harness_name=harness, no_cache=no_cache,
                          on_progress=progress)

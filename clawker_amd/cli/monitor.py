@@ -52,18 +52,15 @@ def gpus_cmd(ctx: Ctx, fmt):
     if fmt == "json":
         f.io.print(json.dumps(rows, indent=1))
         return
-    from rich.table import Table
-    t = Table(box=None, pad_edge=False)
-    for c in ("GPU", "RENDER", "VRAM", "BUSY", "USED", "POWER", "TEMP", "OWNER"):
-        t.add_column(c)
-    for r in rows:
-        t.add_row(str(r["index"]), r["render"], f"{r['vram_gb']}G",
-                  f"{r['busy_pct']:.0f}%" if r["busy_pct"] is not None else "-",
-                  f"{r['vram_used_gb']}G" if r["vram_used_gb"] is not None else "-",
-                  f"{r['power_w']:.0f}W" if r["power_w"] is not None else "-",
-                  f"{r['temp_c']:.0f}C" if r["temp_c"] else "-",
-                  r["owner"] or "-")
-    f.io.print(t)
+    from ..tui.components import plain_table
+    f.io.print(plain_table(
+        ("GPU", "RENDER", "VRAM", "BUSY", "USED", "POWER", "TEMP", "OWNER"),
+        [(r["index"], r["render"], f"{r['vram_gb']}G",
+          f"{r['busy_pct']:.0f}%" if r["busy_pct"] is not None else None,
+          f"{r['vram_used_gb']}G" if r["vram_used_gb"] is not None else None,
+          f"{r['power_w']:.0f}W" if r["power_w"] is not None else None,
+          f"{r['temp_c']:.0f}C" if r["temp_c"] else None,
+          r["owner"]) for r in rows]))
 
 
 @cli.command("stats")

@@ -48,6 +48,14 @@ class IOStreams:
         except (AttributeError, ValueError):
             return False
 
+    def is_stderr_tty(self) -> bool:
+        if self._force_tty is not None:
+            return self._force_tty
+        try:
+            return self.stderr.isatty()
+        except (AttributeError, ValueError):
+            return False
+
     def can_prompt(self) -> bool:
         if os.environ.get("CLAWKER_NO_PROMPT") or os.environ.get("CI"):
             return False

@@ -73,3 +73,29 @@ def test_exporter_event_counters(isolated_env):
     log2.emit("cp_ready")
     c.update()
     assert 'clawker_cp_events_total{event="cp_ready"} 2' in "\n".join(c.lines())
+
+
+def test_tui_components_render():
+    """Shared component library: progress tree marks, kv panel, table
+    (reference: internal/tui reusable components)."""
+    from rich.console import Console
+    import io as _io
+    from clawker_amd.tui.components import ProgressSteps, kv_panel, plain_table
+    steps = ProgressSteps(["base", "harness"])
+    steps.start("base")
+    steps.add("seed configs", parent="harness")
+    steps.done("base", detail="cached")
+    steps.fail("seed configs", detail="exit 1")
+    steps.skip("harness")
+    assert steps.failed
+    con = Console(file=_io.StringIO(), force_terminal=False, width=80)
+    con.print(steps)
+    out = con.file.getvalue()
+    assert "✓ base" in out and "cached" in out
+    assert "✗ seed configs" in out and "- harness" in out
+    con = Console(file=_io.StringIO(), force_terminal=False, width=80)
+    con.print(kv_panel("sandbox", {"gpus": [0, 1], "fw": True, "err": None}))
+    con.print(plain_table(("A", "B"), [(1, None), ("x", ["y", "z"])]))
+    out = con.file.getvalue()
+    assert "sandbox" in out and "0, 1" in out and "yes" in out
+    assert "y, z" in out
