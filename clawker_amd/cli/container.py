@@ -320,8 +320,8 @@ def ps_cmd(ctx: Ctx, show_all, quiet, fmt, filters):
                 label_filters[kv[1]] = kv[2]
     infos = f.engine().list(project=project, all_states=True,
                             label_filters=label_filters or None)
-    if not show_all:
-        infos = [i for i in infos if i.state == "running"]
+    if not show_all:    # paused sandboxes stay visible (docker parity)
+        infos = [i for i in infos if i.state in ("running", "paused")]
     if quiet:
         for i in infos:
             f.io.print(i.name)
@@ -368,10 +368,12 @@ def exec_cmd(ctx: Ctx, user, workdir, env_kv, interactive, tty, name, cmd):
         if code != 0:
             raise ExitError(code)
         return
+    # docker semantics: only -i consumes stdin. Slurping whenever stdin
+    # is merely non-tty blocks forever under a parent that keeps the pipe
+    # open without writing (observed: exec inside a CI bash wrapper).
     stdin = b""
-    if interactive or not sys.stdin.isatty():
-        if not sys.stdin.isatty():
-            stdin = sys.stdin.buffer.read()
+    if interactive and not sys.stdin.isatty():
+        stdin = sys.stdin.buffer.read()
     with f.engine().client(sb) as c:
         stage = {"argv": list(cmd)}
         if user:

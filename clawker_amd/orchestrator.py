@@ -28,7 +28,12 @@ def _clawkerignore_filter(ws_root: Path):
     """shutil.copytree ignore callable from .clawkerignore (reference:
     .clawkerignore excluding paths from snapshot workspaces). Patterns are
     fnmatch-style against names and workspace-relative paths; a trailing
-    '/' means directories only; '#' comments."""
+    '/' means directories only; '#' comments.
+
+    Always excluded, pattern or not: the clawker data/runtime/state dirs
+    when they happen to live INSIDE the workspace — a snapshot copy that
+    descended into its own destination volume would otherwise self-copy
+    recursively until ENAMETOOLONG."""
     import fnmatch
     patterns: list[tuple[str, bool]] = []
     try:
@@ -40,13 +45,23 @@ def _clawkerignore_filter(ws_root: Path):
             patterns.append((line.rstrip("/"), dir_only))
     except OSError:
         pass
-    if not patterns:
-        return None
+    own_dirs = set()
+    for d in (consts.data_dir(), consts.runtime_dir(), consts.state_dir()):
+        try:
+            own_dirs.add(Path(d).resolve())
+        except OSError:
+            pass
 
     def ignore(dirpath, names):
         rel_dir = Path(dirpath).resolve().relative_to(ws_root.resolve())
         out = set()
         for n in names:
+            try:
+                if (Path(dirpath) / n).resolve() in own_dirs:
+                    out.add(n)
+                    continue
+            except OSError:
+                pass
             rel = str(rel_dir / n) if str(rel_dir) != "." else n
             is_dir = (Path(dirpath) / n).is_dir()
             for pat, dir_only in patterns:

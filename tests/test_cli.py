@@ -454,3 +454,41 @@ def test_monitor_extensions_lane(proj, tmp_path):
     assert r.exit_code == 0
     r = _invoke(["skill", "list"])
     assert r.exit_code == 0
+
+
+def test_exec_stdin_semantics(isolated_env, tmp_path, monkeypatch):
+    """Without -i, exec must NOT consume stdin (a parent holding the pipe
+    open would block it forever); with -i, piped stdin reaches the cmd."""
+    import subprocess
+    ws = tmp_path / "exproj"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text("project: extest\n")
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    orch = Orchestrator(load_config(ws))
+    name = "clawker.extest.a"
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO) + os.pathsep + env.get("PYTHONPATH", "")
+    try:
+        orch.run(RunOptions(agent="a", name=name, autostart=True,
+                            firewall=False, host_services=False,
+                            cmd=["/bin/sleep", "30"]))
+        # no -i, stdin = a pipe that NEVER closes: exec must still finish
+        r, w = os.pipe()
+        p = subprocess.run(
+            [sys.executable, "-m", "clawker_amd", "exec", name, "--",
+             "/bin/echo", "done-no-stdin"],
+            stdin=r, capture_output=True, timeout=20, env=env,
+            cwd=ws)
+        os.close(r), os.close(w)
+        assert p.returncode == 0 and b"done-no-stdin" in p.stdout, p.stderr
+        # -i: piped stdin is delivered
+        p = subprocess.run(
+            [sys.executable, "-m", "clawker_amd", "exec", "-i", name, "--",
+             "/bin/cat"],
+            input=b"piped-bytes", capture_output=True, timeout=20, env=env,
+            cwd=ws)
+        assert p.returncode == 0 and b"piped-bytes" in p.stdout, p.stderr
+    finally:
+        orch.teardown(name, force=True)
+        orch.close()

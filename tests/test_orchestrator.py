@@ -142,3 +142,33 @@ def test_clawkerignore_respected_in_snapshots(orch, proj):
     assert "keep.txt" in out and "data.txt" in out
     assert ".git" not in out and "api.secret" not in out and "build" not in out
     orch.teardown(name, force=True)
+
+
+def test_snapshot_excludes_nested_clawker_dirs(isolated_env, tmp_path, monkeypatch):
+    """A data dir nested INSIDE the workspace must not be snapshot-copied
+    (regression: self-copy recursion until ENAMETOOLONG)."""
+    import os
+    ws = tmp_path / "nested"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text("project: nested\n")
+    (ws / "code.txt").write_text("x")
+    # relocate the clawker dirs INSIDE the workspace (plausible user setup)
+    for var, sub in (("CLAWKER_DATA_DIR", "d"), ("CLAWKER_STATE_DIR", "s")):
+        monkeypatch.setenv(var, str(ws / sub))
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    orch = Orchestrator(load_config(ws))
+    name = "clawker.nested.a"
+    try:
+        orch.run(RunOptions(agent="a", name=name, autostart=True,
+                            firewall=False, host_services=False,
+                            workspace_mode="snapshot",
+                            cmd=["/bin/sh", "-c", "ls; test ! -e d && test ! -e s"]))
+        assert orch.engine.wait(name, timeout_s=30) == 0, \
+            orch.engine.logs(name).decode()
+        vol = orch.engine.ensure_volume(f"{name}-snapshot", {})[0]
+        assert (vol / "code.txt").is_file()
+        assert not (vol / "d").exists() and not (vol / "s").exists()
+    finally:
+        orch.teardown(name, force=True)
+        orch.close()
