@@ -239,3 +239,18 @@ def test_pause_unpause_freezes_workload(engine, tmp_path):
     engine.start(info.name)
     assert engine.inspect(info.name).state == "running"
     engine.remove(info.name, force=True)
+
+
+@requires_isolation
+def test_stop_thaws_paused_sandbox(engine):
+    """stop on a paused sandbox unfreezes then terminates (no stranded
+    SIGSTOPped processes)."""
+    info = engine.create(_spec("pzstop", "trap 'exit 7' TERM; while true; do sleep 0.1; done"))
+    engine.start(info.name)
+    time.sleep(0.3)
+    engine.pause(info.name)
+    assert engine.inspect(info.name).state == "paused"
+    code = engine.stop(info.name, timeout_s=8)
+    assert code == 7, code         # the TERM trap ran -> tree was thawed
+    assert engine.inspect(info.name).state == "exited"
+    engine.remove(info.name, force=True)
