@@ -35,6 +35,17 @@ def volume_ls(ctx: Ctx, fmt):
     f.io.print(t)
 
 
+@volume_group.command("create")
+@click.argument("name")
+@pass_factory
+def volume_create(ctx: Ctx, name):
+    """Create a named volume (mount with -v NAME:/dst)."""
+    f = ctx.factory
+    full = name if name.startswith("clawker.") else f"clawker.user.{name}"
+    path, fresh = f.engine().ensure_volume(full, {})
+    f.io.print(full if fresh else f"{full} (exists)")
+
+
 @volume_group.command("rm")
 @click.argument("names", nargs=-1, required=True)
 @pass_factory

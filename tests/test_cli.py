@@ -492,3 +492,19 @@ def test_exec_stdin_semantics(isolated_env, tmp_path, monkeypatch):
     finally:
         orch.teardown(name, force=True)
         orch.close()
+
+
+def test_firewall_rotate_ca_and_volume_create(proj):
+    _invoke(["init", "--yes", "--name", "rc", "--harness", "echo"])
+    from clawker_amd.firewall import mitm
+    ca1, _ = mitm.ensure_ca()
+    pem1 = ca1.read_bytes()
+    r = _invoke(["firewall", "rotate-ca"])
+    assert r.exit_code == 0, r.output
+    assert mitm.ensure_ca()[0].read_bytes() != pem1     # new CA material
+    r = _invoke(["volume", "create", "scratch"])
+    assert r.exit_code == 0 and "clawker.user.scratch" in r.output
+    r = _invoke(["volume", "ls"])
+    assert "scratch" in r.output
+    r = _invoke(["firewall", "refresh"])                # reload alias
+    assert r.exit_code == 0
