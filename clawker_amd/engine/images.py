@@ -262,11 +262,20 @@ class ImageStore:
                 raise ConflictError(f"{in_path}: not a clawker image tarball")
             # refuse traversal: every member stays under layers/ or is the
             # manifest (absolute paths and .. segments rejected)
-            for n in names:
+            for m in tar.getmembers():
+                n = m.name
                 parts = Path(n).parts
                 if (n != "manifest.json" and parts[:1] != ("layers",)) \
                         or ".." in parts or n.startswith("/"):
                     raise ConflictError(f"{in_path}: unsafe member {n!r}")
+                # sym/hardlinks must stay inside the tree: an absolute or
+                # ..-escaping linkname could alias host paths into the
+                # store (write-through or read exposure)
+                if m.issym() or m.islnk():
+                    ln = m.linkname
+                    if ln.startswith("/") or ".." in Path(ln).parts:
+                        raise ConflictError(
+                            f"{in_path}: unsafe link {n!r} -> {ln!r}")
             manifest = json.loads(tar.extractfile("manifest.json").read())
             if "images" in manifest:      # chain format
                 metas = [ImageMeta.from_dict(d) for d in manifest["images"]]

@@ -241,3 +241,26 @@ def test_container_commit_snapshot_layer(ctx):
     assert "tooled" in logs and "gone" in logs
     eng.remove(name, force=True)
     eng.remove(name2, force=True)
+
+
+def test_image_load_rejects_escaping_links(isolated_env, tmp_path):
+    """Symlink/hardlink members pointing outside the layer tree are
+    refused (write-through / read-exposure hardening)."""
+    import io
+    import tarfile
+    import pytest as _pytest
+    from clawker_amd.engine.images import ImageStore
+    from clawker_amd.errors import ConflictError
+    for linkname, typ in (("/etc", tarfile.SYMTYPE),
+                          ("../../../../etc/shadow", tarfile.LNKTYPE)):
+        evil = tmp_path / "evil-link.tar.gz"
+        with tarfile.open(evil, "w:gz") as tar:
+            m = json.dumps({"name": "evil:latest", "layers": ["x"]}).encode()
+            ti = tarfile.TarInfo("manifest.json"); ti.size = len(m)
+            tar.addfile(ti, io.BytesIO(m))
+            ti = tarfile.TarInfo("layers/x/fs/link")
+            ti.type = typ
+            ti.linkname = linkname
+            tar.addfile(ti)
+        with _pytest.raises(ConflictError, match="unsafe link"):
+            ImageStore().load(evil)
