@@ -65,12 +65,22 @@ def gpus_cmd(ctx: Ctx, fmt):
 
 @cli.command("stats")
 @click.option("--no-stream", is_flag=True, help="print once and exit")
+@click.option("--format", "fmt", default="", help="json for machine output")
 @click.option("--interval", type=float, default=1.0, show_default=True)
 @pass_factory
-def stats_cmd(ctx: Ctx, no_stream, interval):
+def stats_cmd(ctx: Ctx, no_stream, fmt, interval):
     """Live per-sandbox + per-GPU stats (replaces docker stats)."""
     f = ctx.factory
     from ..monitor.stats import collect_stats, render_stats
+    if fmt == "json":
+        import dataclasses
+        snap = collect_stats(f.engine())
+        f.io.print(json.dumps({
+            "gpus": [dataclasses.asdict(g) for g in snap.gpus],
+            "allocations": snap.allocations,
+            "sandboxes": snap.sandboxes,
+        }, indent=1, default=str))
+        return
     if no_stream or not f.io.is_stdout_tty():
         snap = collect_stats(f.engine())
         f.io.print(render_stats(snap))

@@ -99,3 +99,14 @@ def test_tui_components_render():
     out = con.file.getvalue()
     assert "sandbox" in out and "0, 1" in out and "yes" in out
     assert "y, z" in out
+
+
+def test_stats_json_format(isolated_env):
+    import json as _json
+    from click.testing import CliRunner
+    from clawker_amd.cli.root import cli
+    r = CliRunner().invoke(cli, ["stats", "--format", "json"])
+    if r.exception is not None and not isinstance(r.exception, SystemExit):
+        raise r.exception
+    d = _json.loads(r.output)
+    assert set(d) == {"gpus", "allocations", "sandboxes"}
