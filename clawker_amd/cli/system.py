@@ -84,3 +84,35 @@ def system_prune(ctx: Ctx, with_volumes):
                 eng.remove_volume(v["name"])
                 removed["volumes"] += 1
     f.io.success(f"pruned: {removed}")
+
+
+@system_group.command("info")
+@pass_factory
+def system_info(ctx: Ctx):
+    """Host + engine summary (docker info analog)."""
+    import platform
+    from .. import __version__, consts
+    from ..engine.engine import detect_backend, native_bin_dir
+    f = ctx.factory
+    eng = f.engine()
+    infos = eng.list()
+    from ..gpu import GPUInventory
+    inv = GPUInventory.detect()
+    from ..controlplane.client import CPClient
+    out = {
+        "version": __version__,
+        "backend": detect_backend(),
+        "kernel": platform.release(),
+        "native_bin": str(native_bin_dir()),
+        "data_dir": str(consts.data_dir()),
+        "runtime_dir": str(consts.runtime_dir()),
+        "sandboxes": {
+            "total": len(infos),
+            "running": sum(1 for i in infos if i.state == "running"),
+            "paused": sum(1 for i in infos if i.state == "paused"),
+        },
+        "images": len(eng.images.list()),
+        "gpus": len(inv),
+        "control_plane": CPClient(auto_start=False).running(),
+    }
+    f.io.print(json.dumps(out, indent=1))

@@ -508,3 +508,11 @@ def test_firewall_rotate_ca_and_volume_create(proj):
     assert "scratch" in r.output
     r = _invoke(["firewall", "refresh"])                # reload alias
     assert r.exit_code == 0
+
+
+def test_system_info(proj):
+    r = _invoke(["system", "info"])
+    assert r.exit_code == 0, r.output
+    d = json.loads(r.output)
+    assert d["backend"] in ("ns", "proc")
+    assert "sandboxes" in d and "paused" in d["sandboxes"]

@@ -140,3 +140,28 @@ def test_hostproxy_reachable_from_inside_sandbox(isolated_env, tmp_path, hostpro
     finally:
         orch.teardown(name, force=True)
         orch.close()
+
+
+def test_hostproxy_crash_respawn(isolated_env):
+    """A killed hostproxy is detected and respawned by ensure_running
+    (failure-detection contract: subsystems recover, never wedge)."""
+    import os
+    import signal as _sig
+    from clawker_amd.hostproxy import HostProxyManager
+    mgr = HostProxyManager()
+    mgr.ensure_running()
+    assert mgr.running()
+    # find and kill the daemon
+    from clawker_amd import consts
+    pid = int((consts.runtime_dir() / "hostproxy.pid").read_text())
+    os.kill(pid, _sig.SIGKILL)
+    # the killed daemon lingers as a zombie of this test process, so
+    # liveness is judged by health, not kill(pid, 0)
+    deadline = time.time() + 5
+    while time.time() < deadline and mgr._healthy():
+        time.sleep(0.02)
+    assert not mgr._healthy()
+    mgr.ensure_running()          # respawn, same socket path
+    status, _ = mgr.request("GET", "/healthz")
+    assert status == 200
+    mgr.stop()
