@@ -156,6 +156,14 @@ def test_agent_ready_gate(engine):
     # init marker persisted in the upper layer
     marker = info.statedir / "upper" / "var/lib/clawker/initialized"
     assert marker.exists()
+    # ...and survives a restart: hello reports initialized=True so the
+    # CP's InitPlan runs exactly once per sandbox (reference: clawkerd
+    # Hello handler + AgentInitialized marker, session.go:421)
+    engine.start(info.name)
+    with engine.client(info.name) as c:
+        assert c.hello()["initialized"] is True
+        c.agent_ready()
+    engine.wait(info.name)
 
 
 def test_duplicate_create_conflicts(engine):
