@@ -86,3 +86,20 @@ def test_claude_harness_floor_golden(isolated_env):
                    for r in h.egress)
     check_golden("claude_egress_floor.txt", "\n".join(floor) + "\n")
     assert any("claude.ai" in f and "/share" in f for f in floor)
+
+
+def test_config_json_schemas_generated():
+    """gen_docs emits JSON schemas for clawker.yaml + settings.yaml
+    (reference: cmd/gen-docs JSON-schema generation)."""
+    import json
+    from pathlib import Path
+    repo = Path(__file__).resolve().parent.parent
+    for name, top in (("clawker", "agent"), ("settings", "control_plane")):
+        sch = json.loads((repo / "docs/schema" / f"{name}.schema.json").read_text())
+        assert sch["$schema"].startswith("http://json-schema.org/")
+        assert top in sch["properties"]
+        assert sch["properties"][top]["type"] == "object"
+    # nested defaults survive: gpu.count default present in project schema
+    sch = json.loads((repo / "docs/schema/clawker.schema.json").read_text())
+    gpu = sch["properties"]["gpu"]["properties"]
+    assert "count" in gpu
