@@ -82,7 +82,11 @@ def init_cmd(ctx: Ctx, name, preset, harness, gpus, vcs, yes):
     if add_domains:
         doc.setdefault("security", {})["egress"] = [
             {"dst": d, "proto": "tls", "port": 443} for d in add_domains]
-    cfg_path.write_text(yaml.safe_dump(doc, sort_keys=False))
+    # $schema header: editor validation via the generated JSON schema
+    # (reference: gen-docs schema stamping, ARCHITECTURE.md:160-265)
+    header = ("# yaml-language-server: $schema="
+              "https://clawker-amd.local/schema/clawker.schema.json\n")
+    cfg_path.write_text(header + yaml.safe_dump(doc, sort_keys=False))
     ignore = root / consts.IGNORE_FILE_NAME
     if not ignore.exists():
         ignore.write_text("# paths excluded from snapshot workspaces\n.git/\n")
