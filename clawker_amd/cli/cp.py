@@ -57,10 +57,8 @@ def cp_agents(ctx: Ctx):
 @pass_factory
 def cp_events(ctx: Ctx, n, follow):
     cp = ctx.factory.controlplane()
-    seen = 0
     for ev in cp.events(n):
         ctx.factory.io.print(json.dumps(ev))
-        seen = max(seen, int(ev.get("ts", 0) * 1e6))
     if not follow:
         return
     # live push stream via cpd pub/sub (reference: Topic subscriber)
