@@ -98,3 +98,21 @@ def test_native_sources_compile_warning_clean():
     r = subprocess.run(["make", "-n", "native"], cwd=REPO,
                        capture_output=True, text=True)
     assert r.returncode == 0, r.stderr
+
+
+def test_admin_ops_client_daemon_coverage():
+    """Every op the CPClient can send is handled by cpd, and cpd's
+    handled set is what the client + CLI rely on (reference:
+    TestAdminMethodScopes_CoversAllRPCs — no RPC falls through)."""
+    daemon_src = (REPO / "clawker_amd/controlplane/daemon.py").read_text()
+    handled = set(re.findall(r'op == "([a-z_]+)"', daemon_src)) | \
+        set(re.findall(r'req\.get\("op"\) == "([a-z_]+)"', daemon_src))
+    client_src = (REPO / "clawker_amd/controlplane/client.py").read_text()
+    sent = set(re.findall(r'"op": "([a-z_]+)"', client_src))
+    missing = sent - handled
+    assert missing == set(), f"client sends unhandled ops: {missing}"
+    # the firewall attach path used by the orchestrator is present
+    for required in ("fw_attach", "fw_detach", "fw_add_rules", "fw_status",
+                     "events", "events_follow", "bypass", "reload_policy",
+                     "agents", "status", "ping", "shutdown"):
+        assert required in handled, f"cpd lost op {required}"
