@@ -293,3 +293,45 @@ def test_multiple_attach_clients_see_console(orch):
         c1.close()
         c2.close()
         orch.engine.stop(name)
+
+
+@requires_isolation
+def test_ckd_survives_malformed_frames(isolated_env):
+    """PID-1 resilience: garbage and hostile frames never take ckd down
+    (reference: panic-recovery-everywhere contract, recover.go:32)."""
+    import socket
+    import struct
+    from clawker_amd import consts
+    from clawker_amd.engine import Engine, SandboxSpec
+    from clawker_amd.engine import wire
+    eng = Engine()
+    name = consts.SANDBOX_NAME_PREFIX + "rb.frames"
+    eng.create(SandboxSpec(name=name, hostname="x", autostart=True, netns=True,
+                           cmd=["/bin/sleep", "30"]))
+    eng.start(name)
+    sock_path = eng.ctl_sock(name)
+    try:
+        # 1: valid frame with invalid JSON -> error frame, conn usable
+        s = wire.connect_unix(sock_path)
+        body = b"{not json"
+        s.sendall(struct.pack(">I", len(body)) + body)
+        resp = wire.recv_frame(s)
+        assert resp["t"] == "error"
+        wire.send_frame(s, {"t": "hello"})
+        assert wire.recv_frame(s)["t"] == "hello"
+        s.close()
+        # 2: oversized length prefix -> dropped, daemon alive
+        s = wire.connect_unix(sock_path)
+        s.sendall(struct.pack(">I", 1 << 30))
+        assert wire.recv_frame(s) is None      # ckd closed us
+        s.close()
+        # 3: raw garbage (no framing) -> dropped eventually, daemon alive
+        s = wire.connect_unix(sock_path)
+        s.sendall(b"\xff" * 64)
+        s.close()
+        # daemon still fully functional
+        code, out, _ = eng.exec(name, ["/bin/echo", "alive"])
+        assert code == 0 and b"alive" in out
+    finally:
+        eng.remove(name, force=True)
+        eng.close()
