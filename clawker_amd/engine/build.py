@@ -111,10 +111,23 @@ def commit_sandbox(engine: Engine, name: str, image_name: str,
     upper = Path(info.statedir) / "upper"
     if not upper.is_dir():
         raise EngineError("commit", f"no writable layer for {name}")
+    # freeze a running workload while the layer is copied so the snapshot
+    # is point-in-time consistent (docker commit --pause semantics)
+    froze = False
+    if info.state == "running":
+        try:
+            engine.pause(name)
+            froze = True
+        except Exception:
+            pass
     tmp_id, layer_fs = engine.images.new_layer_dir()
-    r = subprocess.run(
-        ["cp", "-a", "--reflink=auto", f"{upper}/.", str(layer_fs)],
-        capture_output=True, text=True)
+    try:
+        r = subprocess.run(
+            ["cp", "-a", "--reflink=auto", f"{upper}/.", str(layer_fs)],
+            capture_output=True, text=True)
+    finally:
+        if froze:
+            engine.unpause(name)
     if r.returncode != 0:
         shutil.rmtree(layer_fs.parent, ignore_errors=True)
         raise EngineError("commit", f"layer copy failed: {r.stderr.strip()}")
