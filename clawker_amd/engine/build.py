@@ -60,6 +60,14 @@ def build_layer(engine: Engine, base_image: str, script: str,
             info = engine.inspect(name)
             if info.state != "running":
                 code = info.exit_code
+                if code is None:
+                    # the shim writes exit.json right AFTER the child is
+                    # reaped: a poll can land in that window. wait() holds
+                    # through the grace loop for the definitive code.
+                    try:
+                        code = engine.wait(name, timeout_s=5)
+                    except EngineError:
+                        code = -1
                 break
             if time.monotonic() > deadline:
                 engine.stop(name, timeout_s=3)
