@@ -638,8 +638,10 @@ def cp_cmd(ctx: Ctx, src, dst):
         buf = io.BytesIO()
         with tarfile.open(fileobj=buf, mode="w") as tf:
             tf.add(str(sp), arcname=sp.name)
+        import shlex
+        q = shlex.quote(dst_path)
         code, out, err = f.engine().exec(
-            dst_sb, ["/bin/sh", "-c", f"mkdir -p {dst_path} && tar -C {dst_path} -xf -"],
+            dst_sb, ["/bin/sh", "-c", f"mkdir -p {q} && tar -C {q} -xf -"],
             stdin=buf.getvalue())
         if code != 0:
             raise ClawkerError(f"tar extract failed: {err.decode()[-200:]}")

@@ -39,7 +39,9 @@ class Step:
 
 
 def _exists_in_sandbox(client: CkdClient, path: str) -> bool:
-    code, _, _ = client.exec([{"argv": ["/bin/sh", "-c", f"test -e {path}"]}])
+    import shlex
+    code, _, _ = client.exec(
+        [{"argv": ["/bin/sh", "-c", f"test -e {shlex.quote(path)}"]}])
     return code == 0
 
 
