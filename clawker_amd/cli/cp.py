@@ -45,10 +45,17 @@ def cp_status(ctx: Ctx):
 
 
 @cp_group.command("agents")
+@click.option("-a", "--all", "show_all", is_flag=True,
+              help="include rows for removed sandboxes")
 @pass_factory
-def cp_agents(ctx: Ctx):
+def cp_agents(ctx: Ctx, show_all):
+    """Agent registry (cpd's sqlite; rows close out as 'removed' when
+    their sandbox leaves the engine)."""
     cp = ctx.factory.controlplane()
-    ctx.factory.io.print(json.dumps(cp.agents(), indent=1))
+    rows = cp.agents()
+    if not show_all:
+        rows = [r for r in rows if r.get("state") != "removed"]
+    ctx.factory.io.print(json.dumps(rows, indent=1))
 
 
 @cp_group.command("events")
