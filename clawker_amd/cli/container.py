@@ -355,13 +355,24 @@ container_group.add_command(ps_cmd, "ps")
 @click.option("-e", "--env", "env_kv", multiple=True)
 @click.option("-i", "--interactive", is_flag=True, help="keep stdin open")
 @click.option("-t", "--tty", is_flag=True, help="allocate a pseudo-TTY")
+@click.option("-d", "--detach", is_flag=True, help="run in background")
 @click.argument("name")
 @click.argument("cmd", nargs=-1, required=True, type=click.UNPROCESSED)
 @pass_factory
-def exec_cmd(ctx: Ctx, user, workdir, env_kv, interactive, tty, name, cmd):
+def exec_cmd(ctx: Ctx, user, workdir, env_kv, interactive, tty, detach, name, cmd):
     """Run a command in a running sandbox (-it for an interactive shell)."""
     f = ctx.factory
     sb = resolve_sandbox_name(f, name)
+    if detach:
+        stage = {"argv": list(cmd)}
+        if user:
+            stage["user"] = user
+        if workdir:
+            stage["cwd"] = workdir
+        with f.engine().client(sb) as c:
+            eid = c.exec_start([stage], env=_parse_kv(env_kv) or None)
+        f.io.print(eid)
+        return
     if tty:
         code = _exec_interactive(f, sb, list(cmd), user, workdir,
                                  _parse_kv(env_kv) or None)

@@ -151,6 +151,22 @@ class CkdClient:
             # ignore unrelated events
         return code, bytes(out), bytes(err)
 
+    def exec_start(self, stages: list[dict[str, Any]], stdin: bytes = b"",
+                   env: dict[str, str] | None = None) -> str:
+        """Fire-and-forget exec (docker exec -d): returns the exec id once
+        ckd has spawned it. The job keeps running after this client
+        disconnects (ckd orphans it; output goes to the void)."""
+        CkdClient._exec_seq += 1
+        eid = f"d{CkdClient._exec_seq}-{int(time.time() * 1000) & 0xFFFFFF}"
+        msg: dict[str, Any] = {"t": "exec", "id": eid, "stages": stages}
+        if stdin:
+            msg["stdin"] = wire.b64(stdin)
+        if env:
+            msg["env"] = env
+        self.send(msg)
+        self._wait_for("started")
+        return eid
+
     def exec_start_tty(self, argv: list[str], user: str = "", cwd: str = "",
                        env: dict[str, str] | None = None) -> str:
         """Start an interactive (pty) exec; returns its id. The caller

@@ -516,3 +516,33 @@ def test_system_info(proj):
     d = json.loads(r.output)
     assert d["backend"] in ("ns", "proc")
     assert "sandboxes" in d and "paused" in d["sandboxes"]
+
+
+def test_exec_detach(isolated_env, tmp_path):
+    """exec -d returns immediately with an exec id; the job keeps running
+    after the client disconnects and its side effects land."""
+    ws = tmp_path / "edproj"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text("project: edtest\n")
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    orch = Orchestrator(load_config(ws))
+    name = "clawker.edtest.a"
+    try:
+        orch.run(RunOptions(agent="a", name=name, autostart=True,
+                            firewall=False, host_services=False,
+                            cmd=["/bin/sleep", "30"]))
+        import time as _t
+        r = _invoke(["exec", "-d", name, "--",
+                     "/bin/sh", "-c", "sleep 0.3; echo bg > /tmp/detached"])
+        assert r.exit_code == 0 and r.output.strip().startswith("d")
+        deadline = _t.time() + 10
+        found = False
+        while _t.time() < deadline and not found:
+            code, out, _ = orch.engine.exec(name, ["/bin/cat", "/tmp/detached"])
+            found = code == 0 and b"bg" in out
+            _t.sleep(0.05)
+        assert found
+    finally:
+        orch.teardown(name, force=True)
+        orch.close()
