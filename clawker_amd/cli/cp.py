@@ -80,3 +80,7 @@ def cp_serve():
     import sys
     from ..controlplane.daemon import main as cpd_main
     sys.exit(cpd_main())
+
+
+# docker-style top-level alias: `clawker events [-f]`
+cli.add_command(cp_events, "events")
