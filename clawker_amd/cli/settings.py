@@ -114,3 +114,16 @@ def version_cmd(ctx: Ctx):
     from ..engine.engine import detect_backend
     ctx.factory.io.print(f"clawker-amd {__version__} "
                          f"(isolation backend: {detect_backend()})")
+
+
+@cli.command("completion")
+@click.argument("shell", type=click.Choice(["bash", "zsh", "fish"]))
+@pass_factory
+def completion_cmd(ctx: Ctx, shell):
+    """Print the shell-completion script (eval or source it: e.g.
+    `source <(clawker completion bash)`)."""
+    from click.shell_completion import get_completion_class
+    from .root import cli as root_cli
+    cls = get_completion_class(shell)
+    comp = cls(root_cli, {}, "clawker", "_CLAWKER_COMPLETE")
+    ctx.factory.io.print(comp.source())

@@ -546,3 +546,11 @@ def test_exec_detach(isolated_env, tmp_path):
     finally:
         orch.teardown(name, force=True)
         orch.close()
+
+
+def test_completion_scripts(proj):
+    for shell, marker in (("bash", "_clawker_completion"),
+                          ("zsh", "#compdef clawker"),
+                          ("fish", "complete")):
+        r = _invoke(["completion", shell])
+        assert r.exit_code == 0 and marker in r.output
