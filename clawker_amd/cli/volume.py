@@ -35,6 +35,30 @@ def volume_ls(ctx: Ctx, fmt):
     f.io.print(t)
 
 
+@volume_group.command("inspect")
+@click.argument("names", nargs=-1, required=True)
+@pass_factory
+def volume_inspect(ctx: Ctx, names):
+    """Volume details incl. disk usage (docker volume inspect analog)."""
+    import subprocess
+    from pathlib import Path as _P
+    f = ctx.factory
+    vols = {v["name"]: v for v in f.engine().db.list_volumes()}
+    out = []
+    for n in names:
+        full = n if n.startswith("clawker.") else f"clawker.user.{n}"
+        v = vols.get(full) or vols.get(n)
+        if v is None:
+            from ..errors import NotFoundError
+            raise NotFoundError(f"volume not found: {n}")
+        du = subprocess.run(["du", "-sb", v["path"]], capture_output=True,
+                            text=True)
+        size = int(du.stdout.split()[0]) if du.returncode == 0 else None
+        out.append({**v, "size_bytes": size,
+                    "exists": _P(v["path"]).is_dir()})
+    f.io.print(json.dumps(out, indent=1))
+
+
 @volume_group.command("create")
 @click.argument("name")
 @pass_factory

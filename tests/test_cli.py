@@ -554,3 +554,14 @@ def test_completion_scripts(proj):
                           ("fish", "complete")):
         r = _invoke(["completion", shell])
         assert r.exit_code == 0 and marker in r.output
+
+
+def test_volume_inspect(proj):
+    _invoke(["volume", "create", "vi"])
+    r = _invoke(["volume", "inspect", "vi"])
+    assert r.exit_code == 0, r.output
+    d = json.loads(r.output)[0]
+    assert d["name"] == "clawker.user.vi" and d["exists"] is True
+    assert isinstance(d["size_bytes"], int)
+    r = _invoke(["volume", "inspect", "nope"])
+    assert r.exit_code != 0
