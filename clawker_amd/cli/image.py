@@ -133,6 +133,30 @@ def image_load(ctx: Ctx, rename, tarball):
     f.io.print(f"{meta.name}  layers={len(meta.layers)}")
 
 
+@image_group.command("history")
+@click.argument("name")
+@pass_factory
+def image_history(ctx: Ctx, name):
+    """Layer chain of an image, top-most first (docker history analog)."""
+    import subprocess
+    f = ctx.factory
+    store = f.engine().images
+    from rich.table import Table
+    t = Table(box=None, pad_edge=False)
+    for c in ("IMAGE", "LAYER", "SIZE", "CREATED"):
+        t.add_column(c)
+    now = time.time()
+    for meta in store._image_chain(name):
+        for lid in reversed(meta.layers):
+            du = subprocess.run(["du", "-sh", str(store.layer_path(lid))],
+                                capture_output=True, text=True)
+            size = du.stdout.split()[0] if du.returncode == 0 else "?"
+            t.add_row(meta.name, lid[:12], size,
+                      format_age(now - meta.created) if meta.created else "-")
+    t.add_row("hostfs", "(host rootfs)", "-", "-")
+    f.io.print(t)
+
+
 @image_group.command("prune")
 @pass_factory
 def image_prune(ctx: Ctx):

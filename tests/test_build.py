@@ -264,3 +264,19 @@ def test_image_load_rejects_escaping_links(isolated_env, tmp_path):
             tar.addfile(ti)
         with _pytest.raises(ConflictError, match="unsafe link"):
             ImageStore().load(evil)
+
+
+@requires_isolation
+def test_image_history_shows_chain(ctx):
+    cfg, eng = ctx
+    from click.testing import CliRunner
+    from clawker_amd.cli.root import cli
+    from clawker_amd.engine.build import build_image
+    build_image(eng, "h1:latest", "hostfs", "echo a > /a")
+    build_image(eng, "h2:latest", "h1:latest", "echo b > /b")
+    r = CliRunner().invoke(cli, ["image", "history", "h2:latest"])
+    if r.exception is not None and not isinstance(r.exception, SystemExit):
+        raise r.exception
+    assert r.exit_code == 0, r.output
+    assert "h2:latest" in r.output and "h1:latest" in r.output
+    assert "hostfs" in r.output
