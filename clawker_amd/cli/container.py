@@ -521,14 +521,23 @@ container_group.add_command(attach_cmd, "attach")
 
 
 @cli.command("wait")
+@click.option("--timeout", "timeout_s", type=float, default=None,
+              help="give up after SECONDS (exit 124, like timeout(1))")
 @click.argument("names", nargs=-1, required=True)
 @pass_factory
-def wait_cmd(ctx: Ctx, names):
+def wait_cmd(ctx: Ctx, timeout_s, names):
     """Block until sandboxes exit; print their exit codes."""
+    from ..errors import EngineError
     f = ctx.factory
     last = 0
+    deadline = None if timeout_s is None else time.monotonic() + timeout_s
     for n in names:
-        code = f.engine().wait(resolve_sandbox_name(f, n))
+        remaining = None if deadline is None else deadline - time.monotonic()
+        try:
+            code = f.engine().wait(resolve_sandbox_name(f, n),
+                                   timeout_s=remaining)
+        except EngineError:
+            raise ExitError(124)
         f.io.print(str(code))
         last = code
     if last:

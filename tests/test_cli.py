@@ -565,3 +565,22 @@ def test_volume_inspect(proj):
     assert isinstance(d["size_bytes"], int)
     r = _invoke(["volume", "inspect", "nope"])
     assert r.exit_code != 0
+
+
+def test_wait_timeout_flag(isolated_env, tmp_path):
+    ws = tmp_path / "wproj"
+    ws.mkdir()
+    (ws / ".clawker.yaml").write_text("project: wtest\n")
+    from clawker_amd.config import load_config
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    orch = Orchestrator(load_config(ws))
+    name = "clawker.wtest.a"
+    try:
+        orch.run(RunOptions(agent="a", name=name, autostart=True,
+                            firewall=False, host_services=False,
+                            cmd=["/bin/sleep", "30"]))
+        r = _invoke(["wait", "--timeout", "0.3", name])
+        assert r.exit_code == 124
+    finally:
+        orch.teardown(name, force=True)
+        orch.close()
