@@ -161,6 +161,15 @@ class Orchestrator:
         ws_dst = proj.workspace.mount or "/workspace"
         ws_effective = ws_src        # agent-visible workspace path
         if ws_src is not None:
+            # a worktree's .git FILE references the parent repo's gitdir —
+            # snapshotting it would strand the copy (git broken inside)
+            # and orphan the agent's commits from the branch (reference:
+            # guardWorktreeSnapshot, container_create.go:1972)
+            if ws_mode == "snapshot" and (Path(ws_src) / ".git").is_file():
+                raise ClawkerError(
+                    "snapshot workspace mode is not allowed for git "
+                    "worktrees (the agent's work would be stranded); use "
+                    "bind mode or run without --worktree")
             if ws_mode == "snapshot":
                 vol_name = f"{name}-snapshot"
                 vol_path, fresh = self.engine.ensure_volume(

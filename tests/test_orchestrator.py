@@ -172,3 +172,33 @@ def test_snapshot_excludes_nested_clawker_dirs(isolated_env, tmp_path, monkeypat
     finally:
         orch.teardown(name, force=True)
         orch.close()
+
+
+def test_worktree_snapshot_mode_refused(isolated_env, tmp_path):
+    """Snapshot mode over a git worktree is refused (the copy's .git file
+    would dangle and the agent's commits would be stranded)."""
+    import subprocess
+    repo = tmp_path / "wtrepo"
+    repo.mkdir()
+    (repo / ".clawker.yaml").write_text("project: wtsnap\n")
+    (repo / "f").write_text("x")
+    subprocess.run(["git", "init", "-q", "-b", "main"], cwd=repo, check=True)
+    subprocess.run(["git", "add", "-A"], cwd=repo, check=True)
+    subprocess.run(["git", "-c", "user.email=t@t", "-c", "user.name=t",
+                    "commit", "-q", "-m", "i"], cwd=repo, check=True)
+    wt = tmp_path / "wt"
+    subprocess.run(["git", "worktree", "add", "-q", "-b", "b1", str(wt)],
+                   cwd=repo, check=True)
+    from clawker_amd.config import load_config
+    from clawker_amd.errors import ClawkerError
+    from clawker_amd.orchestrator import Orchestrator, RunOptions
+    orch = Orchestrator(load_config(repo))
+    try:
+        with pytest.raises(ClawkerError, match="worktree"):
+            orch.create(RunOptions(agent="a", workspace=wt,
+                                   workspace_mode="snapshot", firewall=False,
+                                   cmd=["/bin/true"]))
+        # and the failed create leaked nothing
+        assert orch.engine.list() == []
+    finally:
+        orch.close()
