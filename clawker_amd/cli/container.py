@@ -781,6 +781,10 @@ def diff_cmd(ctx: Ctx, name):
         raise ClawkerError(f"no writable layer for {sb} (proc backend?)")
     skip_prefixes = ("etc/hostname", "etc/hosts", "etc/resolv.conf",
                      "var/lib/clawker", "run/", "tmp/")
+    try:
+        lowers = f.engine().images.lowerdirs_for(info.image)
+    except Exception:
+        lowers = ["/"]
     rows = []
     for p in sorted(upper.rglob("*")):
         rel = str(p.relative_to(upper))
@@ -797,6 +801,13 @@ def diff_cmd(ctx: Ctx, name):
         elif p.is_dir():
             continue
         else:
-            rows.append(("C", rel))
+            # A if the path exists in no lower layer, C otherwise
+            # (docker diff A/C/D fidelity)
+            mark = "A"
+            for low in lowers:
+                if (_P(low) / rel).exists():
+                    mark = "C"
+                    break
+            rows.append((mark, rel))
     for mark, rel in rows:
         f.io.print(f"{mark} /{rel}")

@@ -350,7 +350,8 @@ def test_container_diff(proj):
     assert r.returncode == 0, r.stderr
     time.sleep(0.5)
     r = clawker("container", "diff", "df")
-    assert "C /newfile" in r.stdout, r.stdout
+    # /newfile exists in no lower layer -> Added (docker diff fidelity)
+    assert "A /newfile" in r.stdout, r.stdout
     if "/etc/issue" in r.stdout:
         assert "D /etc/issue" in r.stdout
     clawker("rm", "-f", "df")
