@@ -130,25 +130,34 @@ class CkdClient:
         out = bytearray()
         err = bytearray()
         code = -1
-        while True:
-            r = self.recv()
-            if r is None:
-                raise EngineError("ckd exec", "connection closed mid-exec")
-            t = r.get("t")
-            if t == "out" and r.get("id") == eid:
-                data = wire.unb64(r.get("data", ""))
-                if r.get("stream") == 1:
-                    out.extend(data)
-                else:
-                    err.extend(data)
-                if on_output:
-                    on_output(int(r.get("stream", 1)), data)
-            elif t == "done" and r.get("id") == eid:
-                code = int(r.get("code", -1))
-                break
-            elif t == "error":
-                raise EngineError("ckd exec", r.get("msg", "error"))
-            # ignore unrelated events
+        # the socket timeout guards connect/handshake, NOT command
+        # duration: a 30s recv timeout would kill any long-running exec
+        # (agents run long commands constantly). ckd dying still ends the
+        # stream: its socket closes and recv() returns None.
+        prev_to = self.sock.gettimeout()
+        self.sock.settimeout(None)
+        try:
+            while True:
+                r = self.recv()
+                if r is None:
+                    raise EngineError("ckd exec", "connection closed mid-exec")
+                t = r.get("t")
+                if t == "out" and r.get("id") == eid:
+                    data = wire.unb64(r.get("data", ""))
+                    if r.get("stream") == 1:
+                        out.extend(data)
+                    else:
+                        err.extend(data)
+                    if on_output:
+                        on_output(int(r.get("stream", 1)), data)
+                elif t == "done" and r.get("id") == eid:
+                    code = int(r.get("code", -1))
+                    break
+                elif t == "error":
+                    raise EngineError("ckd exec", r.get("msg", "error"))
+                # ignore unrelated events
+        finally:
+            self.sock.settimeout(prev_to)
         return code, bytes(out), bytes(err)
 
     def exec_start(self, stages: list[dict[str, Any]], stdin: bytes = b"",

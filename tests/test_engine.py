@@ -262,3 +262,18 @@ def test_stop_thaws_paused_sandbox(engine):
     assert code == 7, code         # the TERM trap ran -> tree was thawed
     assert engine.inspect(info.name).state == "exited"
     engine.remove(info.name, force=True)
+
+
+@requires_isolation
+def test_exec_outlives_socket_timeout(engine):
+    """A command longer than the client's socket timeout must complete:
+    the timeout guards connect/handshake, not exec duration."""
+    info = engine.create(_spec("slowexec", "sleep 60"))
+    engine.start(info.name)
+    with engine.client(info.name, timeout=1.0) as c:
+        t0 = time.monotonic()
+        code, out, _ = c.exec([{"argv": ["/bin/sh", "-c",
+                                         "sleep 2.5; echo long-ok"]}])
+        assert code == 0 and b"long-ok" in out
+        assert time.monotonic() - t0 >= 2.4
+    engine.remove(info.name, force=True)
