@@ -132,6 +132,9 @@ class CPClient:
         try:
             wire.send_frame(s, {"op": "events_follow"})
             hello = wire.recv_frame(s)
+            # a quiet period must not kill the stream: the timeout guards
+            # connect/handshake only (cpd going away closes the socket)
+            s.settimeout(None)
             if not (hello and hello.get("ok")):
                 raise ClawkerError(
                     f"control plane error: {(hello or {}).get('error')}")
