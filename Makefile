@@ -7,7 +7,7 @@ EXT_SUFFIX := $(shell $(PY) -c "import sysconfig; print(sysconfig.get_config_var
 PYBIND_INC := $(shell $(PY) -c "import pybind11; print(pybind11.get_include())")
 PY_INC := $(shell $(PY) -c "import sysconfig; print(sysconfig.get_paths()['include'])")
 
-.PHONY: all native pymod test clean
+.PHONY: all native pymod test test-gpu bench soak docs clean
 
 all: native pymod
 
@@ -31,8 +31,20 @@ native/bin/ckd: native/ckd/ckd.cpp $(COMMON)
 	@mkdir -p native/bin
 	$(CXX) $(CXXFLAGS) -o $@ native/ckd/ckd.cpp
 
-test: native
+test: native pymod
 	python -m pytest tests/ -x -q -m "not gpu"
+
+test-gpu: native pymod
+	python -m pytest tests/ -x -q -m gpu
+
+bench: native pymod
+	python bench.py
+
+soak: native pymod
+	python tools/soak.py 1000 /tmp/clawker-soak.json
+
+docs:
+	python tools/gen_docs.py
 
 clean:
 	rm -rf native/bin build
