@@ -585,3 +585,19 @@ def test_wait_timeout_flag(isolated_env, tmp_path):
     finally:
         orch.teardown(name, force=True)
         orch.close()
+
+
+def test_every_command_help_renders(proj):
+    """--help must render for every command and group (wiring sweep:
+    a bad decorator or import in any verb fails here, not in the field)."""
+    import click
+    from clawker_amd.cli.root import cli as root
+
+    def walk(cmd, path):
+        r = _invoke(path + ["--help"])
+        assert r.exit_code == 0, (path, r.output)
+        if isinstance(cmd, click.Group):
+            for name, sub in cmd.commands.items():
+                walk(sub, path + [name])
+
+    walk(root, [])
