@@ -68,7 +68,7 @@ def main() -> int:
     for i in range(3):
         name = f"clawker.soak.bg{i}"
         orch.run(RunOptions(agent=f"bg{i}", name=name, autostart=True,
-                            firewall=False, cmd=["sleep", "3600"]))
+                            firewall=False, cmd=["sleep", "infinity"]))
         bg.append(name)
     gpu_holder = None
     if have_gpu:
