@@ -559,7 +559,12 @@ class GatewayManager:
             pass
         for ip in targets[:3]:
             try:
-                return socket.create_connection((ip, port), timeout=10)
+                up = socket.create_connection((ip, port), timeout=10)
+                # the 10s guard is for CONNECT only — a streamed response
+                # (SSE / slow LLM output) may legitimately stall longer.
+                # 300s idle cap keeps dead upstreams from pinning threads.
+                up.settimeout(300)
+                return up
             except OSError:
                 continue
         return None
