@@ -23,7 +23,7 @@ clawker_amd/_native$(EXT_SUFFIX): native/pymod/native.cpp
 	$(CXX) $(CXXFLAGS) -shared -fPIC -I$(PYBIND_INC) -I$(PY_INC) \
 		-o $@ native/pymod/native.cpp
 
-native/bin/ckrt: native/ckrt/ckrt.cpp $(COMMON)
+native/bin/ckrt: native/ckrt/ckrt.cpp native/ckrt/devbpf.hpp $(COMMON)
 	@mkdir -p native/bin
 	$(CXX) $(CXXFLAGS) -o $@ native/ckrt/ckrt.cpp
 

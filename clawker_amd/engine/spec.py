@@ -44,6 +44,11 @@ class SandboxSpec:
     device_allow_only: bool = True
     env: dict[str, str] = field(default_factory=dict)
     user: str = ""             # "", "root", "name", or "uid:gid"
+    # preferred ids when a named user must be materialized into the
+    # overlay upper (ns backend): the workspace owner's ids keep
+    # bind-mounted files writable without idmap mounts
+    uid_hint: int = 0
+    gid_hint: int = 0
     workdir: str = "/"
     cmd: list[str] = field(default_factory=list)
     labels: dict[str, str] = field(default_factory=dict)

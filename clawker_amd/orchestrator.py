@@ -284,6 +284,19 @@ class Orchestrator:
         user = opts.user if opts.user is not None else ""
         if not user and harness is not None:
             user = harness.user
+            if user and backend == "proc":
+                # the proc backend has no private /etc to materialize the
+                # harness's user into; if the host can't resolve it either,
+                # degrade EXPLICITLY to root (ckd refuses silent fallback)
+                from .engine.users import is_named_user
+                if is_named_user(user):
+                    import pwd
+                    try:
+                        pwd.getpwnam(user)
+                    except KeyError:
+                        log.warn("harness_user_unavailable", user=user,
+                                 backend=backend)
+                        user = ""
         env.setdefault(consts.ENV_USER, user or "root")
         default_workdir = str(ws_effective) if ws_src is not None else "/"
         workdir = opts.workdir or proj.agent.workdir or default_workdir
@@ -349,6 +362,17 @@ class Orchestrator:
             if host_ssh_auth_sock():
                 env.setdefault("SSH_AUTH_SOCK", "/run/clawker/ssh-agent.sock")
 
+        # named-user materialization hints: the workspace owner's ids keep
+        # bind-mounted files writable by the in-sandbox agent user
+        uid_hint = gid_hint = 0
+        if ws_src is not None:
+            try:
+                import os as _os2
+                st = _os2.stat(ws_src)
+                uid_hint, gid_hint = st.st_uid, st.st_gid
+            except OSError:
+                pass
+
         restart_policy, _, restart_n = opts.restart.partition(":")
         spec = SandboxSpec(
             name=name,
@@ -366,10 +390,37 @@ class Orchestrator:
             env=env,
             user=user,
             workdir=workdir,
+            uid_hint=uid_hint,
+            gid_hint=gid_hint,
             cmd=list(cmd),
             labels=labels,
         )
-        return self.engine.create(spec, image=image)
+        info = self.engine.create(spec, image=image)
+        # agent-owned storage: config volumes + snapshot/share volumes must
+        # be writable by the (possibly just-materialized) sandbox user
+        uid = int(info.labels.get("dev.clawker.uid", "0") or 0)
+        gid = int(info.labels.get("dev.clawker.gid", "0") or 0)
+        if uid:
+            import os as _os3
+            for m in mounts:
+                src = Path(m.src) if m.src else None
+                if src is None or not src.is_dir():
+                    continue
+                if not str(src).startswith(str(consts.volume_store_dir())):
+                    continue
+                try:
+                    st = _os3.stat(src)
+                    if (st.st_uid, st.st_gid) != (uid, gid):
+                        for dirpath, dirnames, filenames in _os3.walk(src):
+                            _os3.chown(dirpath, uid, gid)
+                            for f in filenames:
+                                try:
+                                    _os3.chown(_os3.path.join(dirpath, f), uid, gid)
+                                except OSError:
+                                    pass
+                except OSError:
+                    pass
+        return info
 
     def start(self, name: str):
         """engine start + firewall gateway enrollment (reference:

@@ -153,8 +153,11 @@ def test_restart_policy_on_failure(orch):
     from clawker_amd.orchestrator import RunOptions
     name = "clawker.rtest.rp"
     # fails twice (marker counts attempts), succeeds on the third
+    # user="root": /run/clawker is the only rundir path that survives a
+    # restart (fresh mount ns each attempt) and it is root-only by design
     opts = RunOptions(
         agent="rp", name=name, autostart=True, restart="on-failure:3",
+        user="root",
         cmd=["/bin/sh", "-c",
              "n=$(cat /run/clawker/attempts 2>/dev/null || echo 0); "
              "n=$((n+1)); echo $n > /run/clawker/attempts; "
