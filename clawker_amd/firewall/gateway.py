@@ -479,7 +479,7 @@ class GatewayManager:
         if cl.isdigit():
             remaining = max(0, int(cl) - skip)
             while remaining > 0:
-                data = f.read(min(65536, remaining))
+                data = f.read(min(262144, remaining))
                 if not data:
                     return
                 if dst:
@@ -488,7 +488,7 @@ class GatewayManager:
             return
         if until_eof:
             while True:
-                data = f.read(65536)
+                data = f.read(262144)
                 if not data:
                     return
                 if dst:
@@ -637,12 +637,16 @@ class GatewayManager:
         done = threading.Event()
 
         def pump(src, dst):
+            # 256 KiB zero-copy-ish relay: recv_into a reused buffer
+            # halves allocator churn vs recv() at bulk rates
+            buf = bytearray(262144)
+            mv = memoryview(buf)
             try:
                 while True:
-                    data = src.recv(65536)
-                    if not data:
+                    n = src.recv_into(mv)
+                    if n == 0:
                         break
-                    dst.sendall(data)
+                    dst.sendall(mv[:n])
             except OSError:
                 pass
             finally:

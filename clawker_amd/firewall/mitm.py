@@ -11,6 +11,7 @@ each rundir (SSL_CERT_FILE / CURL_CA_BUNDLE / NODE_EXTRA_CA_CERTS).
 """
 from __future__ import annotations
 
+import os
 import subprocess
 import threading
 from pathlib import Path
@@ -100,5 +101,16 @@ def combined_trust_bundle() -> Path:
         if system.is_file():
             parts.append(system.read_text())
         parts.append(ca_crt.read_text())
-        bundle.write_text("\n".join(parts))
+        content = "\n".join(parts)
+        # atomic replace: concurrent readers (TLS clients loading the
+        # bundle mid-rewrite) must never see a torn file
+        try:
+            if bundle.read_text() == content:
+                return bundle
+        except OSError:
+            pass
+        tmp = bundle.with_suffix(".tmp")
+        tmp.write_text(content)
+        os.chmod(tmp, 0o644)
+        tmp.replace(bundle)
     return bundle
