@@ -173,3 +173,31 @@ def test_gpu_sos_bundle_captures_inventory(orch, tmp_path):
         inv = json.loads(tar.extractfile("gpu/inventory.json").read())
         assert len(inv) >= 1
         assert inv[0]["vram_total"] >= 250 * 2**30     # 288 GB HBM3E
+
+
+def test_gpu_sandbox_nonroot_torch(orch):
+    """VERDICT r01 #1 done-criterion: a NON-ROOT agent can use its pinned
+    GPU. ns backend: the harness 'agent' user is materialized with GPU
+    group membership; proc backend: numeric uid with /dev/kfd's owning
+    gid as primary group (no passwd entry needed)."""
+    import grp
+    from clawker_amd.orchestrator import RunOptions
+    if orch.engine.backend == "ns":
+        user = "agent"
+    else:
+        kfd_gid = os.stat("/dev/kfd").st_gid
+        user = f"54321:{kfd_gid}"
+    payload = (
+        "import os, torch; assert os.getuid() != 0, 'must not be root'; "
+        "assert torch.cuda.is_available(), 'no GPU as non-root'; "
+        "x = torch.randn(256, 256, device='cuda', requires_grad=True); "
+        "(x @ x).sum().backward(); torch.cuda.synchronize(); "
+        "print('NONROOT_GPU_OK uid=%d' % os.getuid(), flush=True)")
+    name = "clawker.gputest.nonroot"
+    orch.run(RunOptions(agent="nonroot", name=name, gpus=1, autostart=True,
+                        user=user, env={"HOME": "/tmp"},
+                        cmd=["python3", "-c", payload]))
+    code = orch.engine.wait(name, timeout_s=300)
+    logs = orch.engine.logs(name)
+    assert code == 0, logs[-800:]
+    assert b"NONROOT_GPU_OK" in logs
