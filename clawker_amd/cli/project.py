@@ -37,9 +37,12 @@ VCS_EGRESS = {
 @click.option("--gpus", type=int, default=0, help="GPUs per agent sandbox")
 @click.option("--vcs", type=click.Choice(sorted(VCS_EGRESS)), multiple=True,
               help="merge VCS egress domains")
+@click.option("--git-protocol", type=click.Choice(["https", "ssh"]),
+              default="https", show_default=True,
+              help="VCS transport: ssh also merges proto-ssh egress rules")
 @click.option("-y", "--yes", is_flag=True, help="non-interactive")
 @pass_factory
-def init_cmd(ctx: Ctx, name, preset, harness, gpus, vcs, yes):
+def init_cmd(ctx: Ctx, name, preset, harness, gpus, vcs, git_protocol, yes):
     """Initialize a clawker project in the current directory."""
     f = ctx.factory
     root = f.cwd.resolve()
@@ -80,8 +83,15 @@ def init_cmd(ctx: Ctx, name, preset, harness, gpus, vcs, yes):
     for v in vcs:
         add_domains += VCS_EGRESS[v]
     if add_domains:
-        doc.setdefault("security", {})["egress"] = [
-            {"dst": d, "proto": "tls", "port": 443} for d in add_domains]
+        rules = [{"dst": d, "proto": "tls", "port": 443} for d in add_domains]
+        if git_protocol == "ssh":
+            # ssh transport rules for the core VCS hosts (reference:
+            # init.go:108-183 github/gitlab/bitbucket x https/ssh merge);
+            # the in-sandbox side goes through clawker-ssh-proxy
+            for v in vcs:
+                rules.append({"dst": VCS_EGRESS[v][0], "proto": "ssh",
+                              "port": 22})
+        doc.setdefault("security", {})["egress"] = rules
     # $schema header: editor validation via the generated JSON schema
     # (reference: gen-docs schema stamping, ARCHITECTURE.md:160-265)
     header = ("# yaml-language-server: $schema="

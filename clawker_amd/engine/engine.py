@@ -253,7 +253,8 @@ class Engine:
                 shutil.copy2(src, dst)
         # host-service helper scripts (hostproxy clients)
         assets = Path(__file__).resolve().parents[1] / "assets"
-        for helper in ("host-open.sh", "git-credential-clawker"):
+        for helper in ("host-open.sh", "git-credential-clawker",
+                       "clawker-ssh-proxy"):
             src = assets / helper
             if src.is_file():
                 shutil.copy2(src, rundir / "bin" / helper)
