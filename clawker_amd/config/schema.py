@@ -104,6 +104,10 @@ class ControlPlaneSettings:
     auto_start: bool = True
     drain_to_zero: bool = True      # CP self-stops when last agent exits
     drain_grace_s: int = 30
+    # a gated (autostart=False) sandbox whose starting client vanished
+    # gets its Init/Boot plans driven by the CP after this grace
+    # (reference: the CP-side executor model, init_steps.go:67)
+    orphan_grace_s: int = 10
 
 
 @dataclass
