@@ -34,6 +34,36 @@ native/bin/ckd: native/ckd/ckd.cpp $(COMMON)
 test: native pymod
 	python -m pytest tests/ -x -q -m "not gpu"
 
+# ---- sanitizer lane: ASAN+UBSAN builds of the native
+# runtime + minijson fuzzer, exercised by the robustness suite ----------
+ASAN_FLAGS := -O1 -std=c++17 -Wall -Wextra -g -fsanitize=address,undefined \
+	-fno-sanitize-recover=all -fno-omit-frame-pointer
+ASAN_BIN := native/bin-asan
+
+native-asan: $(ASAN_BIN)/ckrt $(ASAN_BIN)/ckd $(ASAN_BIN)/ckgw $(ASAN_BIN)/fuzz_minijson
+
+$(ASAN_BIN)/ckrt: native/ckrt/ckrt.cpp native/ckrt/devbpf.hpp $(COMMON)
+	@mkdir -p $(ASAN_BIN)
+	$(CXX) $(ASAN_FLAGS) -o $@ native/ckrt/ckrt.cpp
+
+$(ASAN_BIN)/ckd: native/ckd/ckd.cpp $(COMMON)
+	@mkdir -p $(ASAN_BIN)
+	$(CXX) $(ASAN_FLAGS) -o $@ native/ckd/ckd.cpp
+
+$(ASAN_BIN)/ckgw: native/ckgw/ckgw.cpp $(COMMON)
+	@mkdir -p $(ASAN_BIN)
+	$(CXX) $(ASAN_FLAGS) -o $@ native/ckgw/ckgw.cpp
+
+$(ASAN_BIN)/fuzz_minijson: native/tests/fuzz_minijson.cpp $(COMMON)
+	@mkdir -p $(ASAN_BIN)
+	$(CXX) $(ASAN_FLAGS) -o $@ native/tests/fuzz_minijson.cpp
+
+test-asan: native-asan
+	$(ASAN_BIN)/fuzz_minijson 200000
+	ASAN_OPTIONS=detect_leaks=0 CLAWKER_NATIVE_BIN=$(abspath $(ASAN_BIN)) \
+		python -m pytest tests/test_robustness.py tests/test_proc_backend.py \
+		tests/test_unprivileged_agent.py -x -q -m "not gpu"
+
 test-gpu: native pymod
 	python -m pytest tests/ -x -q -m gpu
 

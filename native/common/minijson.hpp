@@ -8,6 +8,7 @@
 
 #include <cstdint>
 #include <cstdio>
+#include <cstring>
 #include <map>
 #include <memory>
 #include <stdexcept>
@@ -108,6 +109,9 @@ class Value {
         break;
       case Type::Double:
         snprintf(buf, sizeof buf, "%.17g", d_);
+        // keep the double-ness on round trip: "%.17g" renders integral
+        // values (incl. -0) without '.', which would re-parse as Int
+        if (!strpbrk(buf, ".eEnN")) strcat(buf, ".0");
         out += buf;
         break;
       case Type::Str: dump_str(s_, out); break;
