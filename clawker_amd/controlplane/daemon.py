@@ -173,6 +173,12 @@ class CPDaemon:
         # client-coupled — a `run -d` whose CLI died never booted)
         self._orphan_since: dict[str, float] = {}
         self._driving: set[str] = set()
+        # HBM budget watchdog (env hints are not
+        # enforcement — the CP samples drm fdinfo and kills on breach)
+        from ..monitor.hbm import HBMWatchdog
+        self.hbm = HBMWatchdog(
+            self.engine, self.events,
+            mode=getattr(self.settings.gpu, "hbm_enforce", "kill"))
 
     # ------------------------------------------------------------ registry --
     def _open_registry(self) -> sqlite3.Connection:
@@ -240,6 +246,7 @@ class CPDaemon:
                     if name not in running_names:
                         self.bridges.detach(name)
                 self._maybe_drive_orphans(running)
+                self.hbm.check(running)
                 if running:
                     self._last_agent_seen = time.time()
                 elif (self.settings.control_plane.drain_to_zero and self.ready and

@@ -256,6 +256,9 @@ class Orchestrator:
                 env["CLAWKER_HBM_GB"] = str(hbm)
                 pct = max(1, min(100, round(hbm * 100 / 288)))
                 env["GPU_MAX_ALLOC_PERCENT"] = str(pct)
+                # enforcement is host-side: the CP's HBM watchdog reads
+                # this label and kills at >100% of budget (monitor/hbm.py)
+                opts.labels.setdefault("dev.clawker.hbm_gb", str(hbm))
         # remaining A.1 contract: editor default, terminal capability
         # passthrough, telemetry segmentation attributes
         import os as _os
