@@ -1,0 +1,209 @@
+"""HPACK + HTTP/2 unit tests pinned to RFC 7541 Appendix C vectors,
+plus a socketpair-driven H2Connection round trip."""
+import socket
+import struct
+import threading
+
+from clawker_amd.firewall import h2 as H
+
+
+# ---- Huffman (RFC 7541 Appendix C.4/C.6 strings) ----
+
+def test_huffman_rfc_vectors():
+    vectors = {
+        b"www.example.com": bytes.fromhex("f1e3c2e5f23a6ba0ab90f4ff"),
+        b"no-cache": bytes.fromhex("a8eb10649cbf"),
+        b"custom-key": bytes.fromhex("25a849e95ba97d7f"),
+        b"custom-value": bytes.fromhex("25a849e95bb8e8b4bf"),
+        b"302": bytes.fromhex("6402"),
+        b"private": bytes.fromhex("aec3771a4b"),
+        b"Mon, 21 Oct 2013 20:13:21 GMT":
+            bytes.fromhex("d07abe941054d444a8200595040b8166e082a62d1bff"),
+        b"https://www.example.com":
+            bytes.fromhex("9d29ad171863c78f0b97c8e9ae82ae43d3"),
+        b"gzip": bytes.fromhex("9bd9ab"),
+        b"foo=ASDJKHQKBZXOQWEOPIUAXQWEOIU; max-age=3600; version=1":
+            bytes.fromhex("94e7821dd7f2e6c7b335dfdfcd5b3960"
+                          "d5af27087f3672c1ab270fb5291f9587"
+                          "316065c003ed4ee5b1063d5007"),
+    }
+    for plain, encoded in vectors.items():
+        assert H.huffman_encode(plain) == encoded, plain
+        assert H.huffman_decode(encoded) == plain, plain
+
+
+def test_huffman_roundtrip_all_bytes():
+    data = bytes(range(256)) * 3
+    assert H.huffman_decode(H.huffman_encode(data)) == data
+
+
+# ---- HPACK decode (RFC 7541 Appendix C.3/C.5: request sequences) ----
+
+def test_hpack_request_sequence_plain():
+    d = H.HpackDecoder()
+    # C.3.1 first request
+    h1 = d.decode(bytes.fromhex("828684410f7777772e6578616d706c652e636f6d"))
+    assert h1 == [(":method", "GET"), (":scheme", "http"), (":path", "/"),
+                  (":authority", "www.example.com")]
+    # C.3.2 second request (dynamic table hit)
+    h2_ = d.decode(bytes.fromhex("828684be58086e6f2d6361636865"))
+    assert h2_ == [(":method", "GET"), (":scheme", "http"), (":path", "/"),
+                   (":authority", "www.example.com"),
+                   ("cache-control", "no-cache")]
+    # C.3.3 third request
+    h3 = d.decode(bytes.fromhex(
+        "828785bf400a637573746f6d2d6b65790c637573746f6d2d76616c7565"))
+    assert h3 == [(":method", "GET"), (":scheme", "https"),
+                  (":path", "/index.html"), (":authority", "www.example.com"),
+                  ("custom-key", "custom-value")]
+
+
+def test_hpack_request_sequence_huffman():
+    d = H.HpackDecoder()
+    h1 = d.decode(bytes.fromhex("828684418cf1e3c2e5f23a6ba0ab90f4ff"))
+    assert h1 == [(":method", "GET"), (":scheme", "http"), (":path", "/"),
+                  (":authority", "www.example.com")]
+    h2_ = d.decode(bytes.fromhex("828684be5886a8eb10649cbf"))
+    assert h2_[-1] == ("cache-control", "no-cache")
+    h3 = d.decode(bytes.fromhex(
+        "828785bf408825a849e95ba97d7f8925a849e95bb8e8b4bf"))
+    assert h3[-1] == ("custom-key", "custom-value")
+
+
+def test_hpack_eviction():
+    d = H.HpackDecoder(max_table_size=64)
+    # two adds of ~(10+12+32)=54B entries: second evicts the first
+    d.decode(bytes.fromhex("400a637573746f6d2d6b65790c637573746f6d2d76616c7565"))
+    d.decode(bytes.fromhex("40086e65772d6e616d65096e65772d76616c7565"))
+    assert len(d.dynamic) == 1
+    assert d.dynamic[0] == ("new-name", "new-value")
+
+
+def test_hpack_encode_literal_decodable():
+    headers = [(":status", "200"), ("content-length", "42"),
+               ("x-custom", "v" * 200)]
+    blob = H.hpack_encode_literal(headers)
+    out = H.HpackDecoder().decode(blob)
+    assert out == [(n.lower(), v) for n, v in headers]
+
+
+# ---- frame-level round trip over a socketpair ----
+
+def _client_h2_request(sock, path="/ok", body=b""):
+    """Tiny h2 client: preface, SETTINGS, one request, read response."""
+    sock.sendall(H.PREFACE)
+    sock.sendall(b"\x00\x00\x00" + bytes([H.F_SETTINGS, 0]) + b"\x00" * 4)
+
+    def send_frame(ftype, flags, sid, payload):
+        sock.sendall(struct.pack(">I", len(payload))[1:]
+                     + bytes([ftype, flags]) + struct.pack(">I", sid) + payload)
+
+    hdrs = H.hpack_encode_literal([
+        (":method", "POST" if body else "GET"), (":scheme", "https"),
+        (":authority", "svc.test"), (":path", path)])
+    send_frame(H.F_HEADERS,
+               H.FLAG_END_HEADERS | (0 if body else H.FLAG_END_STREAM),
+               1, hdrs)
+    if body:
+        send_frame(H.F_DATA, H.FLAG_END_STREAM, 1, body)
+
+    buf = bytearray()
+    headers = None
+    data = bytearray()
+    decoder = H.HpackDecoder()
+    while True:
+        while len(buf) < 9:
+            chunk = sock.recv(65536)
+            if not chunk:
+                return headers, bytes(data)
+            buf.extend(chunk)
+        ln = int.from_bytes(buf[:3], "big")
+        ftype, flags = buf[3], buf[4]
+        sid = int.from_bytes(buf[5:9], "big") & 0x7FFFFFFF
+        while len(buf) < 9 + ln:
+            buf.extend(sock.recv(65536))
+        payload = bytes(buf[9:9 + ln])
+        del buf[:9 + ln]
+        if ftype == H.F_SETTINGS and not flags & H.FLAG_ACK:
+            sock.sendall(b"\x00\x00\x00" + bytes([H.F_SETTINGS, H.FLAG_ACK])
+                         + b"\x00" * 4)
+        elif ftype == H.F_HEADERS and sid == 1:
+            headers = decoder.decode(payload)
+        elif ftype == H.F_DATA and sid == 1:
+            data.extend(payload)
+            if flags & H.FLAG_END_STREAM:
+                return headers, bytes(data)
+
+
+def test_h2_connection_request_response():
+    a, b = socket.socketpair()
+    seen = {}
+
+    def handler(headers, body):
+        seen["headers"] = dict(headers)
+        seen["body"] = body
+        return 200, [("content-type", "text/plain")], [b"hello ", b"h2"]
+
+    t = threading.Thread(
+        target=lambda: H.H2Connection(b, handler).serve(), daemon=True)
+    t.start()
+    headers, data = _client_h2_request(a, path="/api/x", body=b"req-body")
+    a.close()
+    assert dict(headers)[":status"] == "200"
+    assert data == b"hello h2"
+    assert seen["headers"][":path"] == "/api/x"
+    assert seen["headers"][":method"] == "POST"
+    assert seen["body"] == b"req-body"
+
+
+def test_h2_connection_large_response_flow_control():
+    """Responses larger than the 65535 default window require honoring
+    client WINDOW_UPDATEs."""
+    a, b = socket.socketpair()
+    blob = bytes(range(256)) * 1024   # 256 KiB > default window
+
+    def handler(headers, body):
+        return 200, [], [blob]
+
+    t = threading.Thread(
+        target=lambda: H.H2Connection(b, handler).serve(), daemon=True)
+    t.start()
+
+    # client that grants window as data arrives
+    a.sendall(H.PREFACE)
+    a.sendall(b"\x00\x00\x00" + bytes([H.F_SETTINGS, 0]) + b"\x00" * 4)
+    hdrs = H.hpack_encode_literal([
+        (":method", "GET"), (":scheme", "https"),
+        (":authority", "svc.test"), (":path", "/blob")])
+    a.sendall(struct.pack(">I", len(hdrs))[1:]
+              + bytes([H.F_HEADERS, H.FLAG_END_HEADERS | H.FLAG_END_STREAM])
+              + struct.pack(">I", 1) + hdrs)
+    buf = bytearray()
+    got = bytearray()
+    done = False
+    while not done:
+        while len(buf) < 9:
+            chunk = a.recv(65536)
+            assert chunk, "server closed early"
+            buf.extend(chunk)
+        ln = int.from_bytes(buf[:3], "big")
+        ftype, flags = buf[3], buf[4]
+        while len(buf) < 9 + ln:
+            buf.extend(a.recv(65536))
+        payload = bytes(buf[9:9 + ln])
+        del buf[:9 + ln]
+        if ftype == H.F_SETTINGS and not flags & H.FLAG_ACK:
+            a.sendall(b"\x00\x00\x00" + bytes([H.F_SETTINGS, H.FLAG_ACK])
+                      + b"\x00" * 4)
+        elif ftype == H.F_DATA:
+            got.extend(payload)
+            if flags & H.FLAG_END_STREAM:
+                done = True
+            elif payload:
+                # grant more window on both levels
+                upd = struct.pack(">I", len(payload))
+                for sid in (0, 1):
+                    a.sendall(b"\x00\x00\x04" + bytes([H.F_WINUP, 0])
+                              + struct.pack(">I", sid) + upd)
+    a.close()
+    assert bytes(got) == blob
