@@ -412,10 +412,24 @@ class Store(t.Generic[T]):
         with open(lock, "w") as lf:
             fcntl.flock(lf, fcntl.LOCK_EX)
             try:
+                # comment-preserving surgery when the file exists
+                # (reference: write.go yaml.Node round trip); verified by
+                # re-parse, falls back to a plain dump when unsafe
+                surgical: str | None = None
+                try:
+                    if path.is_file():
+                        from .yamledit import update_yaml_text
+                        surgical = update_yaml_text(path.read_text(), data)
+                except OSError:
+                    surgical = None
                 fd, tmp = tempfile.mkstemp(dir=str(path.parent), prefix="." + path.name)
                 try:
                     with os.fdopen(fd, "w") as f:
-                        yaml.safe_dump(data, f, sort_keys=False, default_flow_style=False)
+                        if surgical is not None:
+                            f.write(surgical)
+                        else:
+                            yaml.safe_dump(data, f, sort_keys=False,
+                                           default_flow_style=False)
                     os.replace(tmp, path)
                 except BaseException:
                     try:
