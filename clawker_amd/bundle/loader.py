@@ -18,6 +18,7 @@ from .. import consts
 from ..config.schema import EgressRule
 from ..errors import NotFoundError
 from ..storage import materialize
+from .versions import VersionSpec
 
 ASSETS = Path(__file__).resolve().parent / "assets"
 
@@ -38,6 +39,9 @@ class Harness:
     # replaced with the in-sandbox prompt path (fleet --prompt)
     prompt_cmd: list[str] = field(default_factory=list)
     user: str = "agent"
+    # how the harness CLI's version is resolved at build time
+    # (reference: versions.go; "@VERSION@" in install steps expands)
+    version: VersionSpec = field(default_factory=VersionSpec)
     stacks: list[str] = field(default_factory=list, metadata={"merge": "union"})
     install: list[BuildStep] = field(default_factory=list)
     env: dict = field(default_factory=dict)

@@ -98,7 +98,21 @@ class Builder:
     # ------------------------------------------------------------- harness --
     def harness_script(self, harness: Harness) -> str:
         lines = ["export DEBIAN_FRONTEND=noninteractive"]
-        lines.append(_steps_script(harness.install))
+        install_steps = harness.install
+        if any("@VERSION@" in st.run for st in install_steps):
+            from ..bundle.versions import resolve_version
+            from ..config.config import load_settings
+            try:
+                settings = load_settings().get()
+            except Exception:
+                settings = None
+            ver, src = resolve_version(harness.version, settings)
+            lines.append(f"# harness version {ver} (source: {src})")
+            install_steps = [
+                BuildStep(run=st.run.replace("@VERSION@", ver),
+                          best_effort=st.best_effort)
+                for st in install_steps]
+        lines.append(_steps_script(install_steps))
         # plan scripts consumed by controlplane/plans.py
         if harness.post_init:
             lines.append(_emit_file("/etc/clawker/post-init.sh", harness.post_init, "755"))

@@ -85,16 +85,29 @@ def bundle_group():
 
 
 @bundle_group.command("install")
-@click.argument("source", type=click.Path(exists=True, file_okay=False))
+@click.argument("source")
 @click.option("--kind", type=click.Choice(["harnesses", "stacks"]),
               default="harnesses", show_default=True)
+@click.option("--name", default="", help="override the installed name")
+@click.option("--ref", default="", help="git branch/tag (git sources)")
 @pass_factory
-def bundle_install(ctx: Ctx, source, kind):
-    """Install a bundle directory into the user tier
+def bundle_install(ctx: Ctx, source, kind, name, ref):
+    """Install a bundle into the user tier
     (~/.config/clawker/<kind>/<name>); it then resolves above the
-    embedded floor."""
+    embedded floor. SOURCE is a local directory or a git URL
+    (https://, git@, file://, or a local repo path ending in .git) —
+    the fetch pipeline caches by value and strips escaping symlinks
+    (reference: internal/bundle install.go)."""
     from pathlib import Path
+    if (source.startswith(("http://", "https://", "git@", "file://",
+                           "ssh://")) or source.endswith(".git")):
+        from ..bundle.install import install_from_git
+        k, n = install_from_git(source, ref=ref, name=name)
+        ctx.factory.io.success(f"installed {k[:-2]} bundle '{n}' from git")
+        return
     src = Path(source)
+    if not src.is_dir():
+        raise ClawkerError(f"not a directory or git URL: {source}")
     manifest = None
     for cand in ("harness.yaml", "manifest.yaml", "stack.yaml"):
         if (src / cand).is_file():
@@ -113,6 +126,21 @@ def bundle_install(ctx: Ctx, source, kind):
     shutil.copytree(src, dst, ignore=_no_symlinks)
     (dst / ".installed").write_text(str(time.time()))
     ctx.factory.io.success(f"installed {kind[:-2]} bundle '{src.name}' -> {dst}")
+
+
+@bundle_group.command("gc")
+@click.option("--dry-run", is_flag=True)
+@pass_factory
+def bundle_gc(ctx: Ctx, dry_run):
+    """Remove installed bundles + cache entries no registered project
+    declares (reference: bundle GC against declaration roots)."""
+    from ..bundle.install import gc
+    removed = gc(dry_run=dry_run)
+    n = sum(len(v) for v in removed.values())
+    verb = "would remove" if dry_run else "removed"
+    ctx.factory.io.print(f"{verb} {n} item(s): "
+                         + ", ".join(f"{k}={v}" for k, v in removed.items() if v)
+                         if n else "nothing to collect")
 
 
 @bundle_group.command("list")

@@ -153,9 +153,18 @@ class NodeGPUSettings:
 
 
 @dataclass
+class BundlerSettings:
+    """Version-resolver endpoints (reference: bundler/versions.go npm +
+    github-release resolvers). Empty = air-gapped; harness pins apply."""
+    npm_registry: str = ""          # e.g. http://mirror:4873
+    github_api: str = ""            # e.g. https://api.github.com
+
+
+@dataclass
 class Settings:
     """settings.yaml (global layer)."""
     version: int = 1
+    bundler: BundlerSettings = field(default_factory=BundlerSettings)
     control_plane: ControlPlaneSettings = field(default_factory=ControlPlaneSettings)
     firewall: FirewallSettings = field(default_factory=FirewallSettings)
     logging: LoggingSettings = field(default_factory=LoggingSettings)
