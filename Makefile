@@ -11,7 +11,7 @@ PY_INC := $(shell $(PY) -c "import sysconfig; print(sysconfig.get_paths()['inclu
 
 all: native pymod
 
-native: native/bin/ckrt native/bin/ckd native/bin/ckgw
+native: native/bin/ckrt native/bin/ckd native/bin/ckgw native/bin/devbpf_probe
 
 native/bin/ckgw: native/ckgw/ckgw.cpp $(COMMON)
 	@mkdir -p native/bin
@@ -30,6 +30,10 @@ native/bin/ckrt: native/ckrt/ckrt.cpp native/ckrt/devbpf.hpp $(COMMON)
 native/bin/ckd: native/ckd/ckd.cpp $(COMMON)
 	@mkdir -p native/bin
 	$(CXX) $(CXXFLAGS) -o $@ native/ckd/ckd.cpp
+
+native/bin/devbpf_probe: native/tests/devbpf_probe.cpp native/ckrt/devbpf.hpp
+	@mkdir -p native/bin
+	$(CXX) $(CXXFLAGS) -o $@ native/tests/devbpf_probe.cpp
 
 test: native pymod
 	python -m pytest tests/ -x -q -m "not gpu"

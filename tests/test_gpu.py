@@ -201,3 +201,34 @@ def test_gpu_sandbox_nonroot_torch(orch):
     logs = orch.engine.logs(name)
     assert code == 0, logs[-800:]
     assert b"NONROOT_GPU_OK" in logs
+
+
+def test_devbpf_enforcement_on_gpu_host():
+    """The GPU box runs pure cgroup-v2: exercise the hand-emitted
+    BPF_PROG_TYPE_CGROUP_DEVICE allow-list there (VERDICT r01 #3)."""
+    import json
+    import subprocess
+    if not Path("/sys/fs/cgroup/cgroup.controllers").exists():
+        pytest.skip("not a cgroup-v2 host")
+    own = Path("/proc/self/cgroup").read_text().split("0::")[1].strip()
+    for base in (Path("/sys/fs/cgroup/clawker-test"),
+                 Path("/sys/fs/cgroup" + own) / "clawker-test"):
+        try:
+            base.mkdir(parents=True, exist_ok=True)
+            break
+        except OSError:
+            continue
+    else:
+        pytest.skip("no writable cgroup subtree")
+    probe = REPO / "native/bin/devbpf_probe"
+    r = subprocess.run([str(probe), str(base)], capture_output=True,
+                       text=True, timeout=30)
+    try:
+        base.rmdir()
+    except OSError:
+        pass
+    doc = json.loads(r.stdout.strip() or "{}")
+    if r.returncode == 3:
+        pytest.skip(f"bpf unavailable: {doc.get('msg')}")
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert doc["status"] == "enforced"

@@ -338,3 +338,36 @@ def test_ckd_survives_malformed_frames(isolated_env):
     finally:
         eng.remove(name, force=True)
         eng.close()
+
+
+@requires_isolation
+def test_devbpf_cgroup_device_enforcement():
+    """cgroup-v2 device allow-list (VERDICT r01 #3 done-criterion): a
+    device NOT in the allow-list cannot be opened from inside the
+    enforced cgroup even though the node exists. Skips when the host
+    refuses bpf() (restricted CI) or runs cgroup v1."""
+    import json
+    import subprocess
+    from pathlib import Path
+    if not Path("/sys/fs/cgroup/cgroup.controllers").exists():
+        pytest.skip("cgroup v1 host (v1 devices controller covers this)")
+    # same placement logic as ckrt's cg2_base fallback
+    base = Path("/sys/fs/cgroup/clawker-test")
+    try:
+        base.mkdir(exist_ok=True)
+    except OSError:
+        own = Path("/proc/self/cgroup").read_text().split("0::")[1].strip()
+        base = Path("/sys/fs/cgroup" + own) / "clawker-test"
+        base.mkdir(parents=True, exist_ok=True)
+    probe = Path(__file__).resolve().parents[1] / "native/bin/devbpf_probe"
+    r = subprocess.run([str(probe), str(base)], capture_output=True,
+                       text=True, timeout=30)
+    try:
+        base.rmdir()
+    except OSError:
+        pass
+    doc = json.loads(r.stdout.strip() or "{}")
+    if r.returncode == 3:
+        pytest.skip(f"bpf unavailable on this host: {doc.get('msg')}")
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert doc["status"] == "enforced"
