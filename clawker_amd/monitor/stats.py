@@ -88,6 +88,15 @@ def collect_stats(engine: Engine, with_tails: bool = True) -> StatsSnapshot:
         }
         if info.state == "running":
             row.update(cgroup_stats(info.name))
+            # per-sandbox VRAM attribution via drm fdinfo (the same
+            # mechanism the HBM watchdog enforces with — monitor/hbm.py)
+            if info.pid and info.gpus:
+                try:
+                    from .hbm import sandbox_pids, vram_bytes_for_pids
+                    row["vram_bytes"] = vram_bytes_for_pids(
+                        sandbox_pids(info.pid))
+                except Exception:
+                    pass
         if with_tails:
             row["tail"] = _console_tail(info.rundir)
         snap.sandboxes.append(row)
@@ -112,16 +121,22 @@ def render_stats(snap: StatsSnapshot):
     from rich.console import Group
     from rich.table import Table
     t = Table(title="sandboxes", box=None, pad_edge=False)
-    for c in ("NAME", "STATE", "PID", "CPU", "MEM", "PIDS", "GPUS"):
+    for c in ("NAME", "STATE", "PID", "CPU", "MEM", "PIDS", "GPUS",
+              "VRAM", "GPU-BUSY"):
         t.add_column(c)
+    busy_by_idx = {g.index: g.busy_pct for g in snap.gpus}
     for s in snap.sandboxes:
         mem = s.get("mem_bytes")
         cpu = s.get("cpu_pct")
+        vram = s.get("vram_bytes")
+        busy = [busy_by_idx[i] for i in s["gpus"] if i in busy_by_idx]
         t.add_row(s["name"], s["state"], str(s.get("pid") or "-"),
                   f"{cpu:.0f}%" if cpu is not None else "-",
                   f"{mem / 2**20:.0f}M" if mem else "-",
                   str(s.get("pids") or "-"),
-                  ",".join(map(str, s["gpus"])) or "-")
+                  ",".join(map(str, s["gpus"])) or "-",
+                  f"{vram / 2**30:.1f}G" if vram else "-",
+                  f"{max(busy):.0f}%" if busy else "-")
     if not snap.gpus:
         return t
     g = Table(title="GPUs (MI355X)", box=None, pad_edge=False)

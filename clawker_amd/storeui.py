@@ -35,10 +35,38 @@ def leaf_fields(schema: type, prefix: str = "") -> list[str]:
     return out
 
 
+def layer_values(store: Store, path: str) -> list[tuple[str, t.Any]]:
+    """Each layer's raw value for a dotted path (layer browser pane —
+    reference: storeui's per-layer inspection, edit.go:268)."""
+    out: list[tuple[str, t.Any]] = []
+    for layer in store.layers:
+        cur: t.Any = layer.data
+        for part in path.split("."):
+            if not isinstance(cur, dict) or part not in cur:
+                cur = _MISSING
+                break
+            cur = cur[part]
+        if cur is not _MISSING:
+            out.append((layer.name, cur))
+    return out
+
+
+_MISSING = object()
+
+
 def edit_store(store: Store, io: IOStreams, max_rounds: int = 100) -> int:
-    """Interactive loop; returns the number of fields changed."""
+    """Interactive loop; returns the number of fields changed.
+
+    Commands at the selection prompt:
+      <n>       edit field n (value prompt, then optional layer target)
+      /text     filter the field list
+      ?n        show every layer's value for field n
+      (empty)   finish
+    """
     p = Prompter(io)
-    fields = leaf_fields(store.schema)
+    all_fields = leaf_fields(store.schema)
+    fields = all_fields
+    writable = [l.name for l in store.layers if l.writable]
     changed = 0
     for _ in range(max_rounds):
         io.print("")
@@ -46,9 +74,32 @@ def edit_store(store: Store, io: IOStreams, max_rounds: int = 100) -> int:
             val = store.get_path(path)
             owner = store.provenance(path) or "defaults"
             io.print(f" {i:2}) {path} = {val!r}  [dim]({owner})[/dim]")
-        sel = p.string("field number to edit (empty to finish)", "")
-        if not sel.strip():
+        sel = p.string(
+            "field number to edit (/filter, ?n = layers, empty to finish)", "")
+        sel = sel.strip()
+        if not sel:
             break
+        if sel.startswith("/"):
+            pat = sel[1:].strip().lower()
+            fields = ([f for f in all_fields if pat in f.lower()]
+                      if pat else all_fields)
+            if not fields:
+                io.error(f"no fields match '{pat}'")
+                fields = all_fields
+            continue
+        if sel.startswith("?"):
+            try:
+                path = fields[int(sel[1:]) - 1]
+            except (ValueError, IndexError):
+                io.error(f"invalid selection: {sel}")
+                continue
+            vals = layer_values(store, path)
+            io.print(f"[bold]{path}[/bold] by layer "
+                     f"(effective: {store.get_path(path)!r}):")
+            for lname, v in vals or [("(none)", "unset everywhere")]:
+                marker = "*" if store.provenance(path) == lname else " "
+                io.print(f"  {marker} {lname}: {v!r}")
+            continue
         try:
             idx = int(sel) - 1
             path = fields[idx]
@@ -61,7 +112,18 @@ def edit_store(store: Store, io: IOStreams, max_rounds: int = 100) -> int:
         except yaml.YAMLError as e:
             io.error(f"bad value: {e}")
             continue
-        layer = store.set(path, value)
+        # layer targeting (reference: per-field save with layer
+        # targeting): default = provenance auto-routing
+        target = None
+        if len(writable) > 1:
+            chosen = p.string(
+                f"target layer [{'/'.join(writable)}] (empty = auto)", "")
+            if chosen.strip():
+                if chosen.strip() not in writable:
+                    io.error(f"no writable layer '{chosen.strip()}'")
+                    continue
+                target = chosen.strip()
+        layer = store.set(path, value, layer=target)
         store.write()
         io.success(f"{path} = {value!r} (layer: {layer})")
         changed += 1
