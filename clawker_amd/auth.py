@@ -38,7 +38,12 @@ def ensure_auth_material() -> bytes:
         return bytes.fromhex(path.read_text().strip())
     path.parent.mkdir(parents=True, exist_ok=True)
     key = secrets.token_bytes(32)
-    fd = os.open(path, os.O_WRONLY | os.O_CREAT | os.O_EXCL, 0o600)
+    try:
+        fd = os.open(path, os.O_WRONLY | os.O_CREAT | os.O_EXCL, 0o600)
+    except FileExistsError:
+        # concurrent creator won the O_EXCL race: use ITS key (two
+        # divergent root keys would invalidate half the fleet's tokens)
+        return bytes.fromhex(path.read_text().strip())
     try:
         os.write(fd, key.hex().encode())
     finally:

@@ -36,23 +36,34 @@ class CPClient:
 
     def ensure_running(self) -> None:
         """Idempotent daemon spawn + readiness poll (reference:
-        ensureRunning + /healthz poll, bootstrap.go:184-203)."""
+        ensureRunning + /healthz poll, bootstrap.go:184-203). Concurrent
+        callers (8-way fleet/soak) serialize on a spawn flock so exactly
+        one cpd ever starts; the daemon additionally refuses to start
+        over a live sibling."""
         if self.running() and self._ping():
             return
-        env = dict(os.environ)
-        consts.log_dir().mkdir(parents=True, exist_ok=True)
-        logf = open(consts.log_dir() / "cpd.out", "ab")
-        subprocess.Popen(
-            [sys.executable, "-m", "clawker_amd.controlplane.daemon"],
-            stdin=subprocess.DEVNULL, stdout=logf, stderr=logf,
-            start_new_session=True, env=env,
-            cwd=str(Path(__file__).resolve().parents[2]))
-        logf.close()
-        deadline = time.monotonic() + self.timeout
-        while time.monotonic() < deadline:
-            if self._ping():
+        import fcntl
+        rd = consts.runtime_dir()
+        rd.mkdir(parents=True, exist_ok=True)
+        with open(rd / "cpd.spawn.lock", "w") as lockf:
+            fcntl.flock(lockf, fcntl.LOCK_EX)
+            # a racer may have finished the spawn while we waited
+            if self.running() and self._ping():
                 return
-            time.sleep(0.02)
+            env = dict(os.environ)
+            consts.log_dir().mkdir(parents=True, exist_ok=True)
+            logf = open(consts.log_dir() / "cpd.out", "ab")
+            subprocess.Popen(
+                [sys.executable, "-m", "clawker_amd.controlplane.daemon"],
+                stdin=subprocess.DEVNULL, stdout=logf, stderr=logf,
+                start_new_session=True, env=env,
+                cwd=str(Path(__file__).resolve().parents[2]))
+            logf.close()
+            deadline = time.monotonic() + self.timeout
+            while time.monotonic() < deadline:
+                if self._ping():
+                    return
+                time.sleep(0.02)
         raise CPSOSError("control plane failed to become ready",
                          assist=f"check {consts.log_dir() / 'cpd.out'}")
 

@@ -474,6 +474,22 @@ class CPDaemon:
         rd = consts.runtime_dir()
         rd.mkdir(parents=True, exist_ok=True)
         sock_path = admin_sock_path()
+        # never unlink a LIVE sibling's socket (a spawn race would
+        # otherwise leave two daemons, the newer stealing the address):
+        # if someone answers ping, we are redundant — exit cleanly
+        try:
+            probe = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+            probe.settimeout(1.0)
+            probe.connect(str(sock_path))
+            from ..engine import wire as _w
+            _w.send_frame(probe, {"op": "ping"})
+            if (_w.recv_frame(probe) or {}).get("ok"):
+                probe.close()
+                log.info("cpd_already_running")
+                return 0
+            probe.close()
+        except OSError:
+            pass
         listener = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
         try:
             sock_path.unlink(missing_ok=True)

@@ -8,6 +8,7 @@ from __future__ import annotations
 
 import json
 import sqlite3
+import threading
 import time
 from pathlib import Path
 
@@ -43,6 +44,11 @@ class StateDB:
         # threads; python's sqlite3 serializes access internally
         self.db = sqlite3.connect(str(self.path), timeout=10.0,
                                   check_same_thread=False)
+        # sqlite3 serializes single statements, but an execute+commit
+        # PAIR from two threads interleaves ("cannot commit - no
+        # transaction is active" under concurrent creates): every
+        # mutation holds this lock across its transaction
+        self._lock = threading.RLock()
         self.db.execute("PRAGMA journal_mode=WAL")
         self.db.execute("PRAGMA busy_timeout=10000")
         self.db.executescript(_SCHEMA)
@@ -55,13 +61,14 @@ class StateDB:
     def add_sandbox(self, name: str, project: str, agent: str, image: str,
                     labels: dict, gpus: list[int], spec_path: str,
                     rundir: str, statedir: str) -> None:
-        self.db.execute(
-            "INSERT OR REPLACE INTO sandboxes "
-            "(name,project,agent,image,created,labels,gpus,spec_path,rundir,statedir) "
-            "VALUES (?,?,?,?,?,?,?,?,?,?)",
-            (name, project, agent, image, time.time(), json.dumps(labels),
-             json.dumps(gpus), spec_path, rundir, statedir))
-        self.db.commit()
+        with self._lock:
+            self.db.execute(
+                "INSERT OR REPLACE INTO sandboxes "
+                "(name,project,agent,image,created,labels,gpus,spec_path,"
+                "rundir,statedir) VALUES (?,?,?,?,?,?,?,?,?,?)",
+                (name, project, agent, image, time.time(), json.dumps(labels),
+                 json.dumps(gpus), spec_path, rundir, statedir))
+            self.db.commit()
 
     def get_sandbox(self, name: str) -> dict | None:
         cur = self.db.execute("SELECT * FROM sandboxes WHERE name=?", (name,))
@@ -91,15 +98,18 @@ class StateDB:
         return out
 
     def remove_sandbox(self, name: str) -> None:
-        self.db.execute("DELETE FROM sandboxes WHERE name=?", (name,))
-        self.db.commit()
+        with self._lock:
+            self.db.execute("DELETE FROM sandboxes WHERE name=?", (name,))
+            self.db.commit()
 
     # -- volumes ---------------------------------------------------------------
     def add_volume(self, name: str, path: str, labels: dict) -> None:
-        self.db.execute(
-            "INSERT OR REPLACE INTO volumes (name,created,labels,path) VALUES (?,?,?,?)",
-            (name, time.time(), json.dumps(labels), path))
-        self.db.commit()
+        with self._lock:
+            self.db.execute(
+                "INSERT OR REPLACE INTO volumes (name,created,labels,path) "
+                "VALUES (?,?,?,?)",
+                (name, time.time(), json.dumps(labels), path))
+            self.db.commit()
 
     def get_volume(self, name: str) -> dict | None:
         cur = self.db.execute("SELECT * FROM volumes WHERE name=?", (name,))
@@ -122,5 +132,6 @@ class StateDB:
         return out
 
     def remove_volume(self, name: str) -> None:
-        self.db.execute("DELETE FROM volumes WHERE name=?", (name,))
-        self.db.commit()
+        with self._lock:
+            self.db.execute("DELETE FROM volumes WHERE name=?", (name,))
+            self.db.commit()

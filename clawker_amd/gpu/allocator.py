@@ -36,9 +36,12 @@ CREATE TABLE IF NOT EXISTS gpu_alloc (
 class GPUAllocator:
     def __init__(self, db, inventory: GPUInventory | None = None,
                  reserve: list[int] | None = None):
-        """db: engine StateDB (shares its sqlite connection)."""
+        """db: engine StateDB (shares its sqlite connection AND its
+        transaction lock — concurrent allocate/release from fleet
+        threads must not interleave execute/commit pairs)."""
         self._db = db.db
         self._statedb = db
+        self._lock = db._lock
         self.inventory = inventory or GPUInventory.detect()
         self.reserve = set(reserve or [])
         self._db.executescript(_SCHEMA)
@@ -58,12 +61,15 @@ class GPUAllocator:
     def reclaim_stale(self, live_sandboxes: set[str]) -> list[int]:
         """Free allocations whose sandbox no longer exists."""
         freed = []
-        for idx, owner in self.allocations().items():
-            if owner not in live_sandboxes:
-                self._db.execute("DELETE FROM gpu_alloc WHERE gpu_index=?", (idx,))
-                freed.append(idx)
+        with self._lock:
+            for idx, owner in self.allocations().items():
+                if owner not in live_sandboxes:
+                    self._db.execute(
+                        "DELETE FROM gpu_alloc WHERE gpu_index=?", (idx,))
+                    freed.append(idx)
+            if freed:
+                self._db.commit()
         if freed:
-            self._db.commit()
             log.info("gpu_reclaimed", indices=freed)
         return freed
 
@@ -73,7 +79,7 @@ class GPUAllocator:
         """Atomically allocate `count` free GPUs to `sandbox`."""
         if count <= 0:
             return []
-        with self._db:   # transaction
+        with self._lock, self._db:   # transaction
             free = self.free_indices()
             if explicit is not None:
                 missing = [i for i in explicit if i not in free]
@@ -98,10 +104,12 @@ class GPUAllocator:
         return chosen
 
     def release(self, sandbox: str) -> list[int]:
-        cur = self._db.execute("SELECT gpu_index FROM gpu_alloc WHERE sandbox=?", (sandbox,))
-        freed = [int(r[0]) for r in cur.fetchall()]
-        self._db.execute("DELETE FROM gpu_alloc WHERE sandbox=?", (sandbox,))
-        self._db.commit()
+        with self._lock:
+            cur = self._db.execute(
+                "SELECT gpu_index FROM gpu_alloc WHERE sandbox=?", (sandbox,))
+            freed = [int(r[0]) for r in cur.fetchall()]
+            self._db.execute("DELETE FROM gpu_alloc WHERE sandbox=?", (sandbox,))
+            self._db.commit()
         if freed:
             log.info("gpu_released", sandbox=sandbox, indices=freed)
         return freed
