@@ -92,6 +92,56 @@ class IOStreams:
         else:
             yield
 
+    # -- pager / markdown / alt screen (reference: iostreams pager +
+    # markdown render + alt-screen management) -------------------------------
+    def markdown(self, text: str) -> None:
+        """Render markdown (help pages, changelog teasers)."""
+        from rich.markdown import Markdown
+        self.console.print(Markdown(text))
+
+    def page(self, text: str, markdown: bool = False) -> None:
+        """Long output through $PAGER on a tty; plain print otherwise
+        (PAGER= / NO_PAGER disables, like the reference's pager)."""
+        pager = os.environ.get("CLAWKER_PAGER", os.environ.get("PAGER", "less"))
+        if (not self.is_stdout_tty() or not pager
+                or os.environ.get("NO_PAGER")):
+            if markdown:
+                self.markdown(text)
+            else:
+                self.print(text)
+            return
+        rendered = text
+        if markdown:
+            from rich.markdown import Markdown
+            tmp = Console(file=io.StringIO(), force_terminal=True,
+                          width=self.terminal_size()[0])
+            tmp.print(Markdown(text))
+            rendered = tmp.file.getvalue()
+        import subprocess
+        env = dict(os.environ)
+        env.setdefault("LESS", "-FRX")   # quit-if-one-screen, raw colors
+        try:
+            subprocess.run([*pager.split()], input=rendered.encode(),
+                           env=env, stdout=self.stdout)
+        except OSError:
+            self.print(rendered)
+
+    @contextmanager
+    def alt_screen(self):
+        """Full-screen mode with guaranteed restore (the reference
+        tracks alt-screen state so raw-mode restore doesn't strand the
+        terminal)."""
+        if not self.is_stdout_tty():
+            yield
+            return
+        self.stdout.write("\x1b[?1049h\x1b[H")
+        self.stdout.flush()
+        try:
+            yield
+        finally:
+            self.stdout.write("\x1b[?1049l")
+            self.stdout.flush()
+
 
 class TestIOStreams(IOStreams):
     """Buffer trio for tests (reference: iostreams.Test())."""

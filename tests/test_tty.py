@@ -126,3 +126,28 @@ def test_interactive_exec_tty(proj):
             p.kill()
         subprocess.run([sys.executable, "-m", "clawker_amd", "rm", "-f", "xt"],
                        capture_output=True, cwd=str(proj), env=env)
+
+
+def test_iostreams_markdown_and_page_fallback():
+    """Markdown render + pager fallback on non-tty (reference:
+    iostreams pager/markdown surface)."""
+    from clawker_amd.iostreams import TestIOStreams
+    io_ = TestIOStreams()
+    io_.markdown("# Title\n\n- item **bold**\n")
+    assert "Title" in io_.out
+    assert "item" in io_.out
+    io2 = TestIOStreams()
+    io2.page("long output\n" * 5)      # non-tty: plain print, no pager
+    assert io2.out.count("long output") == 5
+    io3 = TestIOStreams()
+    io3.page("# H\n\ntext", markdown=True)
+    assert "text" in io3.out
+
+
+def test_iostreams_alt_screen_noop_on_non_tty():
+    from clawker_amd.iostreams import TestIOStreams
+    io_ = TestIOStreams()
+    with io_.alt_screen():
+        io_.print("inside")
+    assert "\x1b[?1049h" not in io_.out     # no escape codes off-tty
+    assert "inside" in io_.out
