@@ -593,7 +593,12 @@ class Engine:
              label_filters: dict | None = None) -> list[SandboxInfo]:
         out = []
         for row in self.db.list_sandboxes(project=project, label_filters=label_filters):
-            info = self.inspect(row["name"])
+            try:
+                info = self.inspect(row["name"])
+            except NotFoundError:
+                # a concurrent remove won the race between the row scan
+                # and the inspect — the listing just doesn't include it
+                continue
             if not all_states and info.state != "running":
                 continue
             out.append(info)
