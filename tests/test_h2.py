@@ -207,3 +207,100 @@ def test_h2_connection_large_response_flow_control():
                               + struct.pack(">I", sid) + upd)
     a.close()
     assert bytes(got) == blob
+
+
+# ---- adversarial/malformed input hardening ----
+
+def _serve_pair(handler=None):
+    a, b = socket.socketpair()
+    handler = handler or (lambda h, body: (200, [], [b"x"]))
+    conn = H.H2Connection(b, handler)
+    t = threading.Thread(target=lambda: _swallow(conn), daemon=True)
+    t.start()
+    return a, t
+
+
+def _swallow(conn):
+    try:
+        conn.serve()
+    except (H.H2Error, OSError):
+        pass
+    finally:
+        try:
+            conn.sock.close()   # unblock a client mid-sendall
+        except OSError:
+            pass
+
+
+def test_h2_bad_preface_rejected():
+    a, t = _serve_pair()
+    a.sendall(b"GET / HTTP/1.1\r\nHost: x\r\n\r\n" + b"\x00" * 10)
+    t.join(timeout=5)
+    assert not t.is_alive()
+
+
+def test_h2_oversized_frame_rejected():
+    a, t = _serve_pair()
+    a.sendall(H.PREFACE)
+    # 16 MiB-1 declared length
+    a.sendall(b"\xff\xff\xff" + bytes([H.F_DATA, 0]) + struct.pack(">I", 1))
+    t.join(timeout=5)
+    assert not t.is_alive()
+
+
+def test_h2_continuation_for_wrong_stream_rejected():
+    a, t = _serve_pair()
+    a.sendall(H.PREFACE)
+    a.sendall(b"\x00\x00\x00" + bytes([H.F_SETTINGS, 0]) + b"\x00" * 4)
+    hdrs = H.hpack_encode_literal([(":method", "GET"), (":path", "/")])
+    # HEADERS without END_HEADERS on stream 1, then CONTINUATION for 3
+    a.sendall(struct.pack(">I", len(hdrs))[1:] + bytes([H.F_HEADERS, 0])
+              + struct.pack(">I", 1) + hdrs)
+    a.sendall(b"\x00\x00\x02" + bytes([H.F_CONT, H.FLAG_END_HEADERS])
+              + struct.pack(">I", 3) + b"\x00\x00")
+    t.join(timeout=5)
+    assert not t.is_alive()
+
+
+def test_h2_huge_request_body_bounded():
+    """DATA flood past the 64 MiB request cap terminates the session
+    instead of buffering unbounded."""
+    a, t = _serve_pair()
+    a.sendall(H.PREFACE)
+    a.sendall(b"\x00\x00\x00" + bytes([H.F_SETTINGS, 0]) + b"\x00" * 4)
+    hdrs = H.hpack_encode_literal([(":method", "POST"), (":path", "/")])
+    a.sendall(struct.pack(">I", len(hdrs))[1:]
+              + bytes([H.F_HEADERS, H.FLAG_END_HEADERS])
+              + struct.pack(">I", 1) + hdrs)
+    a.settimeout(30)
+
+    def drain():
+        try:
+            while a.recv(65536):
+                pass
+        except OSError:
+            pass
+
+    threading.Thread(target=drain, daemon=True).start()
+    chunk = b"\x00" * 16384
+    try:
+        for _ in range(5000):   # ~80 MiB
+            a.sendall(struct.pack(">I", len(chunk))[1:]
+                      + bytes([H.F_DATA, 0]) + struct.pack(">I", 1) + chunk)
+    except OSError:
+        pass                    # server hung up mid-flood: the point
+    a.close()
+    t.join(timeout=10)
+    assert not t.is_alive()
+
+
+def test_hpack_malformed_inputs_raise_cleanly():
+    import pytest as _pytest
+    d = H.HpackDecoder()
+    for blob in (b"\x80",            # index 0
+                 b"\xff\xff\xff\xff\xff\xff\xff\xff\xff\xff\xff",  # varint bomb
+                 b"\x40\x85abc",     # string past end
+                 b"\xbf",            # huge index
+                 b"\x3f\xff\xff\xff\xff\x7f"):  # table size above cap
+        with _pytest.raises(H.H2Error):
+            d.decode(blob)
