@@ -349,16 +349,33 @@ def test_devbpf_cgroup_device_enforcement():
     import json
     import subprocess
     from pathlib import Path
-    if not Path("/sys/fs/cgroup/cgroup.controllers").exists():
-        pytest.skip("cgroup v1 host (v1 devices controller covers this)")
-    # same placement logic as ckrt's cg2_base fallback
-    base = Path("/sys/fs/cgroup/clawker-test")
-    try:
-        base.mkdir(exist_ok=True)
-    except OSError:
-        own = Path("/proc/self/cgroup").read_text().split("0::")[1].strip()
-        base = Path("/sys/fs/cgroup" + own) / "clawker-test"
-        base.mkdir(parents=True, exist_ok=True)
+    # pure-v2 root, or the hybrid host's unified mount — BPF device
+    # programs attach to any cgroup2 directory
+    if Path("/sys/fs/cgroup/cgroup.controllers").exists():
+        roots = [Path("/sys/fs/cgroup")]
+    elif Path("/sys/fs/cgroup/unified/cgroup.controllers").exists():
+        roots = [Path("/sys/fs/cgroup/unified")]
+    else:
+        pytest.skip("no cgroup2 hierarchy on this host")
+    base = None
+    for root in roots:
+        cand = root / "clawker-test"
+        try:
+            cand.mkdir(exist_ok=True)
+            base = cand
+            break
+        except OSError:
+            try:
+                own = Path("/proc/self/cgroup").read_text().split(
+                    "0::")[1].strip()
+                cand = Path(str(root) + own) / "clawker-test"
+                cand.mkdir(parents=True, exist_ok=True)
+                base = cand
+                break
+            except (OSError, IndexError):
+                continue
+    if base is None:
+        pytest.skip("no writable cgroup2 subtree")
     probe = Path(__file__).resolve().parents[1] / "native/bin/devbpf_probe"
     r = subprocess.run([str(probe), str(base)], capture_output=True,
                        text=True, timeout=30)
