@@ -4,7 +4,9 @@ INSIDE a clawker-amd sandbox (proves the monitoring/profiling north star:
 rocprof works through our /dev/kfd + renderD passthrough), with a host-side
 fallback profile of the identical payload.
 
-Usage: python tools/profile_sandbox.py <output_dir>
+Usage: python tools/profile_sandbox.py <output_dir> [--nonroot]
+  --nonroot: run the profiled workload as an unprivileged uid (r02
+  flagship posture) — proves rocprof + GPU access need no root.
 """
 from __future__ import annotations
 
@@ -33,6 +35,7 @@ PAYLOAD = (
 def main() -> int:
     out_dir = Path(sys.argv[1] if len(sys.argv) > 1 else "gpurun_out/prof").resolve()
     out_dir.mkdir(parents=True, exist_ok=True)
+    nonroot = "--nonroot" in sys.argv
 
     base = Path("/tmp/clawker-prof")
     for var, sub in [
@@ -69,9 +72,19 @@ def main() -> int:
             orch.teardown(name, force=True)
         except Exception:
             pass
+        user = ""
+        env = {}
+        if nonroot:
+            kfd_gid = os.stat("/dev/kfd").st_gid
+            user = ("agent" if orch.engine.backend == "ns"
+                    else f"54321:{kfd_gid}")
+            env = {"HOME": "/tmp"}
+            (out_dir / "sandbox").mkdir(exist_ok=True)
+            os.chmod(out_dir / "sandbox", 0o777)
+            os.chmod(out_dir, 0o777)
         orch.run(RunOptions(
             agent="prof", name=name, gpus=1, gpu_indices=[0], autostart=True,
-            cmd=cmd, firewall=True,
+            cmd=cmd, firewall=True, user=user, env=env,
             mounts=[Mount(src=str(out_dir), dst="/profout")]))
         code = orch.engine.wait(name, timeout_s=240)
         logs = orch.engine.logs(name).decode(errors="replace")
