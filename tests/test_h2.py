@@ -304,3 +304,22 @@ def test_hpack_malformed_inputs_raise_cleanly():
                  b"\x3f\xff\xff\xff\xff\x7f"):  # table size above cap
         with _pytest.raises(H.H2Error):
             d.decode(blob)
+
+
+def test_h2_table_invariants():
+    """ABI-style pins (reference: common.h size asserts mirrored in Go):
+    the HPACK static table and Huffman code table are load-bearing
+    constants — a wrong entry breaks interop with every real client."""
+    assert len(H.STATIC_TABLE) == 61
+    assert len(H._HUFF) == 257                  # 256 symbols + EOS
+    assert H.STATIC_TABLE[0] == (":authority", "")
+    assert H.STATIC_TABLE[1] == (":method", "GET")
+    assert H.STATIC_TABLE[7] == (":status", "200")
+    assert H.STATIC_TABLE[60] == ("www-authenticate", "")
+    # canonical Huffman property: codes are prefix-free (decode map
+    # construction would silently collide otherwise)
+    seen = set()
+    for code, bits in H._HUFF:
+        assert (code, bits) not in seen
+        seen.add((code, bits))
+        assert code < (1 << bits)
