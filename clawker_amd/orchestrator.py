@@ -270,6 +270,14 @@ class Orchestrator:
                 env.setdefault("COLORTERM", _os.environ["COLORTERM"])
         env.setdefault("OTEL_RESOURCE_ATTRIBUTES",
                        f"project={self.cfg.project_slug},agent={opts.agent}")
+        # materialized non-root agents + root-owned parent .git (shared
+        # worktree read-through) trip git's dubious-ownership guard; the
+        # check protects against OTHER users' repos on shared machines —
+        # a threat that cannot exist inside a single-agent sandbox
+        if backend == "ns":
+            env.setdefault("GIT_CONFIG_COUNT", "1")
+            env.setdefault("GIT_CONFIG_KEY_0", "safe.directory")
+            env.setdefault("GIT_CONFIG_VALUE_0", "*")
         if proj.agent.env_file:
             from .dotenv import parse_env_file
             base_dir = self.cfg.project_root or Path.cwd()
@@ -414,11 +422,17 @@ class Orchestrator:
         gid = int(info.labels.get("dev.clawker.gid", "0") or 0)
         if uid:
             import os as _os3
+            # paths CLAWKER created for this sandbox are chowned to the
+            # materialized user: managed volumes + clawker-made worktrees
+            # (a user's own bind-mounted repo is NEVER touched — the
+            # workspace-owner uid hint covers that case instead)
+            own_roots = [str(consts.volume_store_dir()),
+                         str(consts.data_dir() / "worktrees")]
             for m in mounts:
                 src = Path(m.src) if m.src else None
                 if src is None or not src.is_dir():
                     continue
-                if not str(src).startswith(str(consts.volume_store_dir())):
+                if not any(str(src).startswith(r) for r in own_roots):
                     continue
                 try:
                     st = _os3.stat(src)

@@ -170,7 +170,8 @@ def test_cli_cp_copy_roundtrip(proj):
                               capture_output=True, text=True, timeout=timeout,
                               cwd=str(proj), env=env)
 
-    r = clawker("run", "-d", "--agent", "c1", "--no-firewall",
+    # --user root: writes to / (the echo harness default is non-root)
+    r = clawker("run", "-d", "--agent", "c1", "--no-firewall", "-u", "root",
                 "--", "/bin/sh", "-c", "echo sandbox-data > /srcfile; sleep 30")
     assert r.returncode == 0, r.stderr
     time.sleep(0.3)
@@ -344,7 +345,8 @@ def test_container_diff(proj):
                               capture_output=True, text=True, timeout=timeout,
                               cwd=str(proj), env=env)
 
-    r = clawker("run", "-d", "--agent", "df", "--no-firewall",
+    # --user root: mutates / and /etc (the echo harness default is non-root)
+    r = clawker("run", "-d", "--agent", "df", "--no-firewall", "-u", "root",
                 "--no-host-services", "--", "/bin/sh", "-c",
                 "echo x > /newfile; rm /etc/issue 2>/dev/null; sleep 30")
     assert r.returncode == 0, r.stderr

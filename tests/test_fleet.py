@@ -90,7 +90,8 @@ def test_fleet_8way_worktrees_with_firewall(git_proj):
              "python3 -c \"import socket,sys\n"
              "s=socket.socket(); s.settimeout(2)\n"
              "try: s.connect(('203.0.113.9',443)); sys.exit(1)\n"
-             "except OSError: sys.exit(0)\" && echo FW_OK $CLAWKER_AGENT"]))
+             "except OSError: sys.exit(0)\" "
+             "&& echo FW_OK $CLAWKER_AGENT uid=$(id -u)"]))
     assert len(members) == 8
     fleet.wait(members, timeout_s=120)
     failures = [(m.sandbox, m.exit_code) for m in members if m.exit_code != 0]
@@ -98,6 +99,8 @@ def test_fleet_8way_worktrees_with_firewall(git_proj):
     for i, m in enumerate(members):
         logs = fleet.orch.engine.logs(m.sandbox).decode()
         assert f"FW_OK w{i}" in logs
+        # r02: fleet agents run as the materialized non-root user
+        assert "uid=0" not in logs, logs
         # gateway sockets were attached for every member
         rundir = fleet.orch.engine.inspect(m.sandbox).rundir
         assert (rundir / "policy.json").exists()
