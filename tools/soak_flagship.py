@@ -117,8 +117,10 @@ def main() -> int:
             t1 = time.perf_counter()
             code = orch.engine.wait(name, timeout_s=60)
             logs = orch.engine.logs(name).decode()
-            res = json.loads(logs.split("RESULT ", 1)[1].splitlines()[0])
             assert code == 0, logs[-300:]
+            if "RESULT " not in logs:
+                raise AssertionError(f"no RESULT line: {logs[-300:]!r}")
+            res = json.loads(logs.split("RESULT ", 1)[1].splitlines()[0])
             assert res["uid"] != 0 and res["allowed"] == 200
             assert res["denied"] == "blocked"
             assert (ws / f"out-{tag}.json").exists()
