@@ -47,6 +47,35 @@ def run_checks() -> list[dict]:
     v1 = (cg / "memory").exists()
     checks.append(_check("cgroups", v1 or v2, "v2 unified" if v2 else
                          "v1 hybrid" if v1 else "no limits available"))
+    # kernel-side device enforcement: v1 devices controller, or the BPF
+    # device program on a cgroup2 hierarchy (probe does a real
+    # load+attach+deny check — native/tests/devbpf_probe.cpp)
+    dev_v1 = (cg / "devices").exists()
+    cg2_root = (cg if v2 else cg / "unified"
+                if (cg / "unified" / "cgroup.controllers").exists() else None)
+    bpf_detail = ""
+    bpf_ok = False
+    if dev_v1:
+        bpf_ok, bpf_detail = True, "v1 devices controller"
+    elif cg2_root is not None:
+        import subprocess as _sp
+        probe = native_bin_dir() / "devbpf_probe"
+        scratch = cg2_root / "clawker-doctor"
+        try:
+            scratch.mkdir(exist_ok=True)
+            r = _sp.run([str(probe), str(cg2_root)], capture_output=True,
+                        text=True, timeout=15)
+            bpf_ok = r.returncode == 0
+            bpf_detail = ("BPF device program enforced" if bpf_ok else
+                          "bpf() unavailable: /dev construction only")
+        except (OSError, _sp.TimeoutExpired) as e:
+            bpf_detail = f"probe failed: {e}"
+        finally:
+            try:
+                scratch.rmdir()
+            except OSError:
+                pass
+    checks.append(_check("device enforcement", bpf_ok, bpf_detail))
     from ..engine.engine import native_bin_dir
     bins = {b: (native_bin_dir() / b).is_file() for b in ("ckrt", "ckd", "ckgw")}
     checks.append(_check("native runtime", all(bins.values()),
