@@ -66,10 +66,15 @@ def gpus_cmd(ctx: Ctx, fmt):
 @cli.command("stats")
 @click.option("--no-stream", is_flag=True, help="print once and exit")
 @click.option("--format", "fmt", default="", help="json for machine output")
-@click.option("--interval", type=float, default=1.0, show_default=True)
+@click.option("--interval", type=float, default=None,
+              help="refresh seconds (default: settings "
+                   "monitoring.sample_interval_ms)")
 @pass_factory
 def stats_cmd(ctx: Ctx, no_stream, fmt, interval):
     """Live per-sandbox + per-GPU stats (replaces docker stats)."""
+    if interval is None:
+        interval = max(0.1, ctx.factory.config().settings
+                       .monitoring.sample_interval_ms / 1000.0)
     f = ctx.factory
     from ..monitor.stats import collect_stats, render_stats
     if fmt == "json":
@@ -128,11 +133,16 @@ def monitor_status(ctx: Ctx):
 
 
 @cli.command("dashboard")
-@click.option("--interval", type=float, default=1.0, show_default=True)
+@click.option("--interval", type=float, default=None,
+              help="refresh seconds (default: settings "
+                   "monitoring.sample_interval_ms)")
 @pass_factory
 def dashboard_cmd(ctx: Ctx, interval):
     """Full-screen live dashboard: agents x GPUs (reference: tui
     RunDashboard precedent)."""
+    if interval is None:
+        interval = max(0.1, ctx.factory.config().settings
+                       .monitoring.sample_interval_ms / 1000.0)
     f = ctx.factory
     from ..monitor.stats import collect_stats
     from ..tui.dashboard import render_dashboard

@@ -62,8 +62,11 @@ class EventLogger:
         self._emit(logging.ERROR, event, **kv)
 
 
-def setup(log_file: Path | None = None, level: str = "", stderr: bool = False) -> None:
-    """Idempotent global setup: rotating file sink + optional stderr."""
+def setup(log_file: Path | None = None, level: str = "", stderr: bool = False,
+          max_size_mb: int = 10, max_backups: int = 3) -> None:
+    """Idempotent global setup: rotating file sink + optional stderr.
+    Rotation knobs come from settings logging.* (reference: lumberjack
+    rotation, schema.go logging :457)."""
     global _configured
     with _lock:
         if _configured:
@@ -74,7 +77,8 @@ def setup(log_file: Path | None = None, level: str = "", stderr: bool = False) -
         if log_file is not None:
             log_file.parent.mkdir(parents=True, exist_ok=True)
             fh = logging.handlers.RotatingFileHandler(
-                log_file, maxBytes=10 * 1024 * 1024, backupCount=3)
+                log_file, maxBytes=max(1, max_size_mb) * 1024 * 1024,
+                backupCount=max(0, max_backups))
             fh.setFormatter(fmt)
             root.addHandler(fh)
         if stderr or os.environ.get("CLAWKER_LOG_STDERR"):

@@ -254,7 +254,8 @@ class Orchestrator:
             hbm = opts.hbm_gb if opts.hbm_gb is not None else proj.gpu.hbm_gb
             if hbm and hbm > 0:
                 env["CLAWKER_HBM_GB"] = str(hbm)
-                pct = max(1, min(100, round(hbm * 100 / 288)))
+                per_dev = self.cfg.settings.gpu.hbm_gb_per_device or 288
+                pct = max(1, min(100, round(hbm * 100 / per_dev)))
                 env["GPU_MAX_ALLOC_PERCENT"] = str(pct)
                 # enforcement is host-side: the CP's HBM watchdog reads
                 # this label and kills at >100% of budget (monitor/hbm.py)
