@@ -46,7 +46,7 @@ def fw_list(ctx: Ctx, fmt):
 
 @firewall_group.command("add")
 @click.argument("dst")
-@click.option("--proto", type=click.Choice(["tls", "http", "tcp", "udp"]),
+@click.option("--proto", type=click.Choice(["tls", "http", "tcp", "udp", "ssh"]),
               default="tls", show_default=True)
 @click.option("--port", type=int, default=443, show_default=True)
 @click.option("--path", "paths", multiple=True, help="allow-only path prefix ('~' = regex)")
