@@ -65,6 +65,8 @@ def main() -> int:
                                         "denied.test=127.0.0.1")
 
     class H(http.server.BaseHTTPRequestHandler):
+        protocol_version = "HTTP/1.1"   # keep-alive: no reconnect storm
+
         def do_GET(self):
             body = b"OK"
             self.send_response(200)
@@ -75,7 +77,11 @@ def main() -> int:
         def log_message(self, *a):
             pass
 
-    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), H)
+    class Srv(http.server.ThreadingHTTPServer):
+        request_queue_size = 128        # 8-way bursts overflow the
+                                        # default backlog of 5 -> resets
+
+    srv = Srv(("127.0.0.1", 0), H)
     threading.Thread(target=srv.serve_forever, daemon=True).start()
     port = srv.server_address[1]
 
