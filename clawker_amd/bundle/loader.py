@@ -72,14 +72,26 @@ def _search_dirs(kind: str, project_root: Path | None) -> list[Path]:
     return dirs
 
 
+# mtime-keyed parse cache: the claude floor manifest alone costs ~24 ms
+# of PyYAML per load, and the hot path loads a harness twice per sandbox
+# create (orchestrator + firewall floor composition)
+_yaml_cache: dict[tuple, tuple[float, dict]] = {}
+
+
 def _load_yaml(kind: str, name: str, project_root: Path | None) -> dict:
     for d in _search_dirs(kind, project_root):
         for candidate in (d / name / f"{kind[:-2] if kind.endswith('es') else kind}.yaml",
                           d / f"{name}.yaml",
                           d / name / "manifest.yaml"):
             if candidate.is_file():
+                key = (kind, str(candidate))
+                mtime = candidate.stat().st_mtime
+                hit = _yaml_cache.get(key)
+                if hit is not None and hit[0] == mtime:
+                    return dict(hit[1])
                 data = yaml.safe_load(candidate.read_text()) or {}
                 data.setdefault("name", name)
+                _yaml_cache[key] = (mtime, dict(data))
                 return data
     raise NotFoundError(f"{kind[:-2] if kind.endswith('es') else kind} not found: {name}")
 
