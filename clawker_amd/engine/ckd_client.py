@@ -26,12 +26,17 @@ class CkdClient:
         """Connect, retrying until ckd has bound its socket (sandbox boot)."""
         end = time.monotonic() + deadline_s
         last: Exception | None = None
+        # exponential backoff from 0.3 ms: ckd binds ~1-3 ms after the
+        # shim forks, so a fixed 5 ms poll quantized the whole cold
+        # start (headline metric) to its own granularity
+        delay = 0.0003
         while time.monotonic() < end:
             try:
                 return cls(sock_path, timeout=timeout)
             except EngineError as e:
                 last = e
-                time.sleep(0.005)
+                time.sleep(delay)
+                delay = min(delay * 2, 0.005)
         raise EngineError("ckd connect", f"timed out after {deadline_s}s: {last}")
 
     def close(self) -> None:

@@ -391,13 +391,15 @@ class Engine:
         # wait until ckd binds its control socket (or the sandbox dies fast)
         sock = rundir / consts.CKD_SOCK_NAME
         deadline = time.monotonic() + wait_ready_s
+        delay = 0.0003   # backoff: the socket appears ~1-3 ms after fork
         while time.monotonic() < deadline:
             if sock.exists():
                 break
             if (rundir / "exit.json").exists() or proc.poll() is not None:
                 tail = self._tail(rundir / "shim.log")
                 raise EngineError("start", f"sandbox died during boot: {tail}")
-            time.sleep(0.002)
+            time.sleep(delay)
+            delay = min(delay * 2, 0.005)
         else:
             raise EngineError("start", f"timed out waiting for ckd socket ({name})")
         log.info("sandbox_started", sandbox=name, shim_pid=proc.pid)
