@@ -118,6 +118,10 @@ def _metrics_text() -> str:
             lines.append(f"clawker_sandbox_memory_bytes{{{lbl}}} {s['mem_bytes']}")
         if s.get("pids"):
             lines.append(f"clawker_sandbox_pids{{{lbl}}} {s['pids']}")
+        if s.get("vram_bytes") is not None:
+            # per-sandbox drm-fdinfo VRAM attribution (monitor/hbm.py)
+            lines.append(
+                f"clawker_sandbox_vram_bytes{{{lbl}}} {s['vram_bytes']}")
     lines.append(f"clawker_sandboxes_running {running}")
     _event_counters.update()
     lines.extend(_event_counters.lines())
