@@ -423,3 +423,27 @@ def test_readonly_bind_covers_submounts(orch, tmp_path):
         orch.teardown(name, force=True)
     finally:
         subprocess.run(["umount", str(sub)], capture_output=True)
+
+
+@requires_isolation
+def test_console_log_rotation(orch):
+    """A flooding agent must not grow console.log unboundedly (64 MiB
+    rotation to console.log.1 — the docker/lumberjack analog)."""
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.rtest.flood"
+    # ~80 MiB of output: forces one rotation
+    orch.run(RunOptions(
+        agent="flood", name=name, autostart=True, user="root",
+        cmd=["/bin/sh", "-c",
+             "i=0; while [ $i -lt 80 ]; do head -c 1048576 /dev/zero |"
+             " tr '\\0' 'x'; i=$((i+1)); done; echo FLOOD_DONE"]))
+    code = orch.engine.wait(name, timeout_s=120)
+    assert code == 0
+    rundir = orch.engine.inspect(name).rundir
+    main = (rundir / "console.log").stat().st_size
+    rotated = (rundir / "console.log.1")
+    assert rotated.exists(), "no rotation happened"
+    assert main < 70 * 1024 * 1024, f"console.log grew to {main}"
+    logs = orch.engine.logs(name).decode(errors="replace")
+    assert "FLOOD_DONE" in logs
+    orch.teardown(name, force=True)
