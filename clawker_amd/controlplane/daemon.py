@@ -76,10 +76,26 @@ class EventLog:
             self.on_emit(rec)
 
     def tail(self, n: int = 100) -> list[dict]:
+        # read a bounded tail window, not the whole (up to 10 MB) file
         try:
-            lines = self.path.read_text().splitlines()[-n:]
-            return [json.loads(l) for l in lines if l.strip()]
-        except (OSError, ValueError):
+            with open(self.path, "rb") as f:
+                f.seek(0, 2)
+                size = f.tell()
+                window = min(size, max(4096, n * 512))
+                f.seek(size - window)
+                chunk = f.read().decode(errors="replace")
+            lines = chunk.splitlines()
+            if window < size and lines:
+                lines = lines[1:]   # first line may be torn
+            out = []
+            for l in lines[-n:]:
+                if l.strip():
+                    try:
+                        out.append(json.loads(l))
+                    except ValueError:
+                        continue
+            return out
+        except OSError:
             return []
 
 
