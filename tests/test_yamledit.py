@@ -91,3 +91,22 @@ def test_store_set_preserves_comments(tmp_path):
     assert "count: 8" in text
     assert "# two MI355X per sandbox" in text
     assert "hand edited!" in text
+
+
+def test_list_value_updates_keep_comments():
+    """Egress-rule list mutations (the most common `settings set`
+    payload) re-render the list block while sibling comments stay."""
+    doc = """# top
+security:
+  firewall: true   # keep on
+  egress:
+    - {dst: a.test, proto: tls, port: 443}
+"""
+    data = yaml.safe_load(doc)
+    data["security"]["egress"].append(
+        {"dst": "b.test", "proto": "ssh", "port": 22})
+    out = update_yaml_text(doc, data)
+    assert out is not None
+    assert yaml.safe_load(out) == data
+    assert "# keep on" in out
+    assert "# top" in out
