@@ -363,6 +363,14 @@ def exec_cmd(ctx: Ctx, user, workdir, env_kv, interactive, tty, detach, name, cm
     """Run a command in a running sandbox (-it for an interactive shell)."""
     f = ctx.factory
     sb = resolve_sandbox_name(f, name)
+    if not user:
+        # docker semantics: exec runs as the sandbox's configured user
+        # unless -u overrides (plans/admin paths send explicit stages)
+        try:
+            with f.engine().client(sb, timeout=5) as _c:
+                user = _c.hello().get("user", "") or ""
+        except Exception:
+            user = ""
     if detach:
         stage = {"argv": list(cmd)}
         if user:

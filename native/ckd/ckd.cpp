@@ -673,6 +673,9 @@ void handle_frame(Client& cl, const mj::Value& req) {
      // uses clients==1 + !cmd_running to detect a gated sandbox whose
      // starting client died before driving the boot plans
      .set("clients", (int64_t)g_clients.size())
+     // the sandbox's configured user: `exec` defaults to it (docker
+     // semantics); internal plans send explicit stages instead
+     .set("user", g_spec["user"].as_str())
      .set("version", "0.1.0");
     send_to_client(cl, r);
   } else if (t == "agent_ready") {

@@ -284,3 +284,31 @@ def test_volume_chown_never_follows_symlinks(ws_orch, tmp_path):
     assert (st.st_uid, st.st_gid) == (0, 0), \
         "chown followed a symlink out of the snapshot volume!"
     orch.teardown(name, force=True)
+
+
+def test_exec_defaults_to_sandbox_user(ws_orch):
+    """`clawker exec` without -u runs as the sandbox's configured user
+    (docker semantics), never silently root."""
+    import subprocess
+    import sys as _sys
+    orch, ws = ws_orch
+    from clawker_amd.orchestrator import RunOptions
+    name = "clawker.utest.execd"
+    orch.run(RunOptions(agent="execd", name=name, autostart=False,
+                        cmd=["sleep", "60"]))
+    with orch.client(name) as c:
+        c.agent_ready()
+    import os as _os
+    env = dict(_os.environ, PYTHONPATH=str(Path(__file__).parents[1]))
+    r = subprocess.run(
+        [_sys.executable, "-m", "clawker_amd", "exec", name, "--",
+         "id", "-u"], capture_output=True, text=True, env=env, timeout=60)
+    assert r.returncode == 0, r.stderr
+    assert int(r.stdout.strip()) >= 1000, r.stdout
+    r2 = subprocess.run(
+        [_sys.executable, "-m", "clawker_amd", "exec", "-u", "root", name,
+         "--", "id", "-u"], capture_output=True, text=True, env=env,
+        timeout=60)
+    assert r2.returncode == 0, r2.stderr
+    assert r2.stdout.strip() == "0"
+    orch.teardown(name, force=True)
