@@ -50,6 +50,7 @@ def run_checks() -> list[dict]:
     # kernel-side device enforcement: v1 devices controller, or the BPF
     # device program on a cgroup2 hierarchy (probe does a real
     # load+attach+deny check — native/tests/devbpf_probe.cpp)
+    from ..engine.engine import native_bin_dir
     dev_v1 = (cg / "devices").exists()
     cg2_root = (cg if v2 else cg / "unified"
                 if (cg / "unified" / "cgroup.controllers").exists() else None)
