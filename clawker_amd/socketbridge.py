@@ -71,14 +71,20 @@ class SocketBridgeManager:
             bridges = []
             for sock_name, target in pairs:
                 path = rundir / sock_name
-                path.unlink(missing_ok=True)
+                # temp-bind + rename-once-listening: a connect in the
+                # bind->listen window would get ECONNREFUSED (same fix
+                # as the egress gateway sockets)
+                tmp = rundir / f".{sock_name}.tmp"
+                tmp.unlink(missing_ok=True)
                 try:
-                    lst = bind_unix(path)
+                    lst = bind_unix(tmp)
                 except OSError as e:
                     log.warn("bridge_bind_failed", sock=sock_name, err=str(e))
                     continue
-                os.chmod(path, 0o666)
+                os.chmod(tmp, 0o666)
                 lst.listen(16)
+                path.unlink(missing_ok=True)
+                tmp.rename(path)
                 lst.settimeout(0.5)
                 br = _Bridge(listener=lst, target=target)
                 threading.Thread(target=self._accept_loop, args=(br,),
