@@ -69,7 +69,11 @@ def leaf_for(domain: str) -> tuple[Path, Path]:
     d = mitm_dir()
     safe = domain.replace("*", "_wild_").replace("/", "_")
     crt, key = d / f"{safe}.crt", d / f"{safe}.key"
-    with _lock:
+    import fcntl
+    with _lock, open(d / ".mint.lock", "w") as lockf:
+        # cross-PROCESS serialization too: cpd and a CLI minting the
+        # same domain concurrently would race openssl's file writes
+        fcntl.flock(lockf, fcntl.LOCK_EX)
         if crt.is_file() and key.is_file():
             return crt, key
         ca_crt, ca_key = ensure_ca()
