@@ -122,7 +122,33 @@ def _build_opts(f: Factory, agent, image, gpus, gpu_indices, hbm_gb, env_kv,
         opts.workspace = wt.path
         if not opts.agent or opts.agent == "agent":
             opts.agent = wt.safe_name
+    _guard_home_workspace(f, opts, cfg)
     return opts
+
+
+def _guard_home_workspace(f: Factory, opts: RunOptions, cfg) -> None:
+    """Mounting $HOME or / as the workspace hands the agent the whole
+    blast radius; require explicit confirmation (reference: the
+    home-mount safety prompt, container/shared/safety.go)."""
+    from pathlib import Path as _P
+    ws = opts.workspace or (cfg.workspace_path() if cfg.project_root else None)
+    if ws is None:
+        return
+    try:
+        ws = _P(ws).resolve()
+        home = _P.home().resolve()
+    except OSError:
+        return
+    if ws == _P("/") or ws == home or ws in home.parents:
+        if f.io.can_prompt():
+            if f.prompter().confirm(
+                    f"workspace {ws} contains your entire home/root — "
+                    "mount it into the sandbox anyway?", default=False):
+                return
+        raise ClawkerError(
+            f"refusing to mount {ws} as the workspace (it contains your "
+            "home directory); use a project subdirectory, or confirm "
+            "interactively")
 
 
 def _boot_and_wait(f: Factory, name: str, interactive: bool, tty: bool,

@@ -603,3 +603,27 @@ def test_every_command_help_renders(proj):
                 walk(sub, path + [name])
 
     walk(root, [])
+
+
+def test_home_workspace_refused_noninteractive(proj, monkeypatch):
+    """Mounting $HOME (or /) as the workspace is refused without
+    interactive confirmation (reference: safety.go home-mount prompt)."""
+    _invoke(["init", "--yes", "--name", "hometest", "--harness", "echo"])
+    import os as _os
+    env = dict(_os.environ, PYTHONPATH=str(REPO), HOME=str(proj.parent))
+    r = subprocess.run(
+        [sys.executable, "-m", "clawker_amd", "run", "-d", "--agent", "h",
+         "--no-firewall", "--workspace-mode", "bind",
+         "--", "true"],
+        capture_output=True, text=True, timeout=60, cwd=str(proj.parent),
+        env=env)
+    # cwd IS $HOME here and has no project -> workspace would be None;
+    # instead drive via a registered project whose root == HOME
+    (proj.parent / ".clawker.yaml").write_text("project: homeproj\n")
+    r = subprocess.run(
+        [sys.executable, "-m", "clawker_amd", "run", "-d", "--agent", "h2",
+         "--no-firewall", "--", "true"],
+        capture_output=True, text=True, timeout=60, cwd=str(proj.parent),
+        env=env)
+    assert r.returncode != 0
+    assert "refusing to mount" in (r.stderr + r.stdout)
