@@ -103,6 +103,7 @@ def image_rm(ctx: Ctx, force, names):
         f.io.print(n)
 
 
+image_rm.help = image_rm.help or "Remove images (docker rmi)."
 cli.add_command(image_rm, "rmi")   # docker-style top-level alias
 
 

@@ -81,7 +81,7 @@ def prompt_rm(ctx: Ctx, name):
 
 @cli.group("bundle")
 def bundle_group():
-    """Harness/stack bundle management (local-dir install tier)."""
+    """Harness/stack bundles: install (dir or git), list, gc."""
 
 
 @bundle_group.command("install")

@@ -110,6 +110,7 @@ def alias_delete(ctx: Ctx, name):
 @cli.command("version")
 @pass_factory
 def version_cmd(ctx: Ctx):
+    """Print the clawker-amd version."""
     from .. import __version__
     from ..engine.engine import detect_backend
     ctx.factory.io.print(f"clawker-amd {__version__} "
