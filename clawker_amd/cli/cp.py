@@ -17,6 +17,7 @@ def cp_group():
 @cp_group.command("up")
 @pass_factory
 def cp_up(ctx: Ctx):
+    """Start the control-plane daemon (idempotent)."""
     cp = ctx.factory.controlplane()
     cp.ensure_running()
     ctx.factory.io.success("control plane ready")
@@ -25,6 +26,7 @@ def cp_up(ctx: Ctx):
 @cp_group.command("down")
 @pass_factory
 def cp_down(ctx: Ctx):
+    """Stop the control-plane daemon (sandboxes keep running; egress fails closed)."""
     cp = ctx.factory.controlplane()
     cp.auto_start = False
     if cp.stop():
@@ -36,6 +38,7 @@ def cp_down(ctx: Ctx):
 @cp_group.command("status")
 @pass_factory
 def cp_status(ctx: Ctx):
+    """Daemon liveness, bypass state and agent count."""
     cp = ctx.factory.controlplane()
     cp.auto_start = False
     if not cp.running():
@@ -63,6 +66,7 @@ def cp_agents(ctx: Ctx, show_all):
 @click.option("-f", "--follow", is_flag=True)
 @pass_factory
 def cp_events(ctx: Ctx, n, follow):
+    """Tail (or follow) the control-plane event stream."""
     cp = ctx.factory.controlplane()
     for ev in cp.events(n):
         ctx.factory.io.print(json.dumps(ev))

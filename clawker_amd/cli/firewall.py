@@ -26,6 +26,7 @@ def _store() -> EgressRulesStore:
 @click.option("--format", "fmt", default="")
 @pass_factory
 def fw_list(ctx: Ctx, fmt):
+    """Current egress rules with sticky identities."""
     f = ctx.factory
     rules = _store().list()
     idents = IdentityAllocator()
@@ -67,6 +68,7 @@ def fw_add(ctx: Ctx, dst, proto, port, paths, deny_paths):
 @click.argument("dst")
 @pass_factory
 def fw_remove(ctx: Ctx, dst):
+    """Remove all rules for DST and reload gateways."""
     f = ctx.factory
     if not _store().remove(dst):
         raise ClawkerError(f"no rule for: {dst}")
@@ -77,6 +79,7 @@ def fw_remove(ctx: Ctx, dst):
 @firewall_group.command("status")
 @pass_factory
 def fw_status(ctx: Ctx):
+    """Gateway attachments, bypass state, rule count."""
     f = ctx.factory
     rules = _store().list()
     infos = f.engine().list()
