@@ -66,7 +66,8 @@ test-asan: native-asan
 	$(ASAN_BIN)/fuzz_minijson 200000
 	ASAN_OPTIONS=detect_leaks=0 CLAWKER_NATIVE_BIN=$(abspath $(ASAN_BIN)) \
 		python -m pytest tests/test_robustness.py tests/test_proc_backend.py \
-		tests/test_unprivileged_agent.py -x -q -m "not gpu"
+		tests/test_unprivileged_agent.py tests/test_firewall_e2e.py \
+		tests/test_ssh_egress.py -x -q -m "not gpu"
 
 test-gpu: native pymod
 	python -m pytest tests/ -x -q -m gpu
