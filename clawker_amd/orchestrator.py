@@ -246,7 +246,7 @@ class Orchestrator:
             consts.ENV_AGENT: opts.agent,
             consts.ENV_WORKSPACE_MODE: ws_mode,
             consts.ENV_WORKSPACE_SOURCE: str(ws_src or ""),
-            consts.ENV_VERSION: "0.1.0",
+            consts.ENV_VERSION: "0.2.0",
             consts.ENV_FIREWALL: "1" if effective_firewall else "0",
         }
         if gpu_indices:

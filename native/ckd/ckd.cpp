@@ -676,7 +676,7 @@ void handle_frame(Client& cl, const mj::Value& req) {
      // the sandbox's configured user: `exec` defaults to it (docker
      // semantics); internal plans send explicit stages instead
      .set("user", g_spec["user"].as_str())
-     .set("version", "0.1.0");
+     .set("version", "0.2.0");
     send_to_client(cl, r);
   } else if (t == "agent_ready") {
     // CP releases the user CMD (reference: boot_steps.go AgentReady)
