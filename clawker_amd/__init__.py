@@ -20,4 +20,4 @@ Go CLI + Docker + eBPF) designed for one dedicated rootful
   reference's `docker stats` loop and compose monitoring stack).
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
