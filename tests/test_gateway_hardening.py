@@ -298,3 +298,33 @@ def test_request_smuggling_guard(gw):
         c.close()
     srv.close()
     assert any(e.get("reason") == "smuggling-guard" for e in events)
+
+
+def test_policy_hot_reload_applies_to_open_sessions(gw):
+    """A rule change lands on the NEXT request of an already-open
+    keep-alive session (PolicyView mtime reload — the reference's
+    Envoy config reload semantics)."""
+    import os as _os
+    import time as _t
+    mgr, rundir, events = gw
+    srv, port = _upstream_once()
+    _policy(rundir, [{"dst": "pinned.test", "proto": "http", "port": port,
+                      "paths": ["/"]}])
+    c = _connect(rundir)
+    c.sendall(f"GET http://pinned.test:{port}/a HTTP/1.1\r\n"
+              f"Host: pinned.test:{port}\r\n\r\n".encode())
+    assert b"200" in c.recv(65536).split(b"\r\n")[0]
+    # tighten the policy: only /allowed remains
+    _t.sleep(0.01)
+    _policy(rundir, [{"dst": "pinned.test", "proto": "http", "port": port,
+                      "paths": ["/allowed"]}])
+    _os.utime(rundir / "policy.json")
+    c.sendall(f"GET http://pinned.test:{port}/a HTTP/1.1\r\n"
+              f"Host: pinned.test:{port}\r\n\r\n".encode())
+    resp = c.recv(65536)
+    assert b"403" in resp.split(b"\r\n")[0], resp[:100]
+    c.sendall(f"GET http://pinned.test:{port}/allowed HTTP/1.1\r\n"
+              f"Host: pinned.test:{port}\r\n\r\n".encode())
+    assert b"200" in c.recv(65536).split(b"\r\n")[0]
+    c.close()
+    srv.close()
