@@ -242,6 +242,8 @@ class CPDaemon:
         driven by the starting client — CLI/fleet — which holds the ckd
         session; the watcher only reconciles host-side attachments.)"""
         while not self._stop.is_set():
+            running = []
+            reconcile_ok = True
             try:
                 infos = self.engine.list()
                 running = [i for i in infos if i.state in ("running", "paused")]
@@ -293,7 +295,20 @@ class CPDaemon:
                     self.events.emit("cp_drain_to_zero")
                     self._stop.set()
             except Exception as e:
+                reconcile_ok = False
                 log.error("watcher_unavailable", err=str(e))
+            # drain-to-zero must run even when a reconcile step throws
+            # (observed: a deleted state dir made every tick raise, so a
+            # test-spawned cpd lived forever); an erroring engine with no
+            # confirmed agents counts as idle
+            if running:
+                self._last_agent_seen = time.time()
+            elif (self.settings.control_plane.drain_to_zero and self.ready
+                  and time.time() - self._last_agent_seen >
+                  self.settings.control_plane.drain_grace_s):
+                log.info("drain_to_zero", reconcile_ok=reconcile_ok)
+                self.events.emit("cp_drain_to_zero")
+                self._stop.set()
             self._stop.wait(1.0)
 
     _last_prune = 0.0
