@@ -277,20 +277,24 @@ def serve() -> int:
     while not stop.is_set():
         stop.wait(2.0)
         # self-exit when no sandboxes remain (reference: 0 containers + 60s)
+        n = 0
         try:
             from ..engine.state import StateDB
             db = StateDB()
             n = len(db.list_sandboxes())
             db.close()
-            if n == 0:
-                idle_since = idle_since or time.time()
-                if time.time() - idle_since > 60:
-                    log.info("hostproxy_idle_exit")
-                    break
-            else:
-                idle_since = None
         except Exception:
-            pass
+            # an erroring/deleted state dir counts as idle — otherwise a
+            # test-spawned daemon lives forever (same bug class as the
+            # cpd drain fix)
+            n = 0
+        if n == 0:
+            idle_since = idle_since or time.time()
+            if time.time() - idle_since > 60:
+                log.info("hostproxy_idle_exit")
+                break
+        else:
+            idle_since = None
     unix_srv.shutdown()
     if tcp_srv:
         tcp_srv.shutdown()
