@@ -259,6 +259,11 @@ def serve() -> int:
     callbacks = CallbackStore()
     handler = type("Handler", (HostProxyHandlerBase,), {"callbacks": callbacks})
 
+    # pid BEFORE the socket becomes healthy: ensure_running returns on
+    # health, and a caller checking running() (pid + socket) in that
+    # instant must not see a half-registered daemon
+    pid_path().parent.mkdir(parents=True, exist_ok=True)
+    pid_path().write_text(str(os.getpid()))
     unix_srv = UnixHTTPServer(str(sock_path()), handler)
     threading.Thread(target=unix_srv.serve_forever, daemon=True).start()
     tcp_srv = None
@@ -268,7 +273,6 @@ def serve() -> int:
     except OSError as e:
         log.warn("hostproxy_tcp_unavailable", err=str(e))
 
-    pid_path().write_text(str(os.getpid()))
     log.info("hostproxy_ready", sock=str(sock_path()))
     stop = threading.Event()
     signal.signal(signal.SIGTERM, lambda *a: stop.set())
