@@ -376,3 +376,48 @@ def test_h2_trailers_do_not_replace_request_headers():
     a.close()
     assert seen["headers"][":path"] == "/api/grpc"
     assert seen["body"] == b"grpc-payload"
+
+
+def test_h2_random_frame_fuzz():
+    """Deterministic random frame streams after a valid preface: the
+    server thread must always terminate (H2Error/OSError) and never
+    hang or raise anything else."""
+    import random
+    rng = random.Random(0xC1A4)
+    for trial in range(40):
+        a, b = socket.socketpair()
+        conn = H.H2Connection(b, lambda h, body: (200, [], [b"x"]))
+        errs = []
+
+        def serve():
+            try:
+                conn.serve()
+            except (H.H2Error, OSError):
+                pass
+            except Exception as e:  # noqa: BLE001
+                errs.append(e)
+            finally:
+                try:
+                    conn.sock.close()
+                except OSError:
+                    pass
+
+        t = threading.Thread(target=serve, daemon=True)
+        t.start()
+        a.settimeout(5)
+        try:
+            a.sendall(H.PREFACE)
+            for _ in range(rng.randrange(1, 12)):
+                ln = rng.randrange(0, 64)
+                ftype = rng.randrange(0, 12)
+                flags = rng.randrange(0, 256)
+                sid = rng.randrange(0, 8)
+                a.sendall(ln.to_bytes(3, "big") + bytes([ftype, flags])
+                          + sid.to_bytes(4, "big") + bytes(
+                              rng.randrange(256) for _ in range(ln)))
+        except OSError:
+            pass
+        a.close()
+        t.join(timeout=5)
+        assert not t.is_alive(), f"h2 server hung on trial {trial}"
+        assert not errs, f"unexpected exception: {errs}"
