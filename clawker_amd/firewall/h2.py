@@ -310,6 +310,9 @@ FLAG_PRIORITY = 0x20
 DEFAULT_WINDOW = 65535
 
 
+MAX_HEADER_BLOCK = 1 << 20    # 1 MiB of (compressed) header bytes
+
+
 class H2Stream:
     def __init__(self, sid: int, send_window: int):
         self.sid = sid
@@ -454,6 +457,8 @@ class H2Connection:
             st = H2Stream(sid, self.peer_initial_window)
             self.streams[sid] = st
         st.header_block.extend(frag)
+        if len(st.header_block) > MAX_HEADER_BLOCK:
+            raise H2Error("header block too large")
         if flags & FLAG_END_STREAM:
             st.end_stream = True
         if flags & FLAG_END_HEADERS:
@@ -475,6 +480,8 @@ class H2Connection:
         if st is None or self._expect_cont != sid:
             raise H2Error("CONTINUATION for unknown stream")
         st.header_block.extend(payload)
+        if len(st.header_block) > MAX_HEADER_BLOCK:
+            raise H2Error("header block too large (CONTINUATION flood)")
         if flags & FLAG_END_HEADERS:
             self._expect_cont = None
             st.end_headers = True

@@ -421,3 +421,25 @@ def test_h2_random_frame_fuzz():
         t.join(timeout=5)
         assert not t.is_alive(), f"h2 server hung on trial {trial}"
         assert not errs, f"unexpected exception: {errs}"
+
+
+def test_h2_continuation_flood_bounded():
+    """Endless CONTINUATION frames must terminate the session at the
+    1 MiB header-block cap, not grow memory."""
+    a, t = _serve_pair()
+    a.settimeout(10)
+    a.sendall(H.PREFACE)
+    hdr_frag = H.hpack_encode_literal([("x", "y" * 200)])
+    # HEADERS without END_HEADERS, then a flood of CONTINUATIONs
+    a.sendall(struct.pack(">I", len(hdr_frag))[1:] + bytes([H.F_HEADERS, 0])
+              + struct.pack(">I", 1) + hdr_frag)
+    try:
+        for _ in range(20000):
+            a.sendall(struct.pack(">I", len(hdr_frag))[1:]
+                      + bytes([H.F_CONT, 0]) + struct.pack(">I", 1)
+                      + hdr_frag)
+    except OSError:
+        pass    # server hung up at the cap: the point
+    a.close()
+    t.join(timeout=10)
+    assert not t.is_alive()
