@@ -441,6 +441,10 @@ class H2Connection:
         pos = 0
         pad = 0
         if flags & FLAG_PADDED:
+            if not payload:
+                # fuzz-found: PADDED flag on an empty payload raised
+                # IndexError out of the session handler
+                raise H2Error("padded frame with empty payload")
             pad = payload[0]
             pos = 1
         if priority and flags & FLAG_PRIORITY:

@@ -380,7 +380,13 @@ def test_revocation_severs_live_tunnel(gw, monkeypatch):
     c.sendall(f"CONNECT pinned.test:{port} HTTP/1.1\r\n\r\n".encode())
     resp = c.recv(65536)
     assert b"200" in resp.split(b"\r\n")[0]
-    assert c.recv(16) == b"HELLO"
+    try:
+        first = c.recv(16)
+    except OSError as e:
+        raise AssertionError(
+            f"relay never delivered: {e}; events={events[-6:]} "
+            f"tunnels={len(mgr._tunnels)}")
+    assert first == b"HELLO"
     # revoke the rule; the sweep must sever the tunnel. The sweep
     # thread's FIRST tick still uses the default 5 s period (the
     # override lands after the thread entered its wait), so allow
