@@ -383,9 +383,16 @@ def test_revocation_severs_live_tunnel(gw, monkeypatch):
     try:
         first = c.recv(16)
     except OSError as e:
+        import sys as _sys
+        import traceback as _tb
+        stacks = []
+        for tid, frame in _sys._current_frames().items():
+            stack = "".join(_tb.format_stack(frame)[-3:])
+            if "pump" in stack or "splice" in stack or "recv" in stack:
+                stacks.append(f"--- thread {tid} ---\n{stack}")
         raise AssertionError(
             f"relay never delivered: {e}; events={events[-6:]} "
-            f"tunnels={len(mgr._tunnels)}")
+            f"tunnels={len(mgr._tunnels)}\n" + "\n".join(stacks[:6]))
     assert first == b"HELLO"
     # revoke the rule; the sweep must sever the tunnel. The sweep
     # thread's FIRST tick still uses the default 5 s period (the
