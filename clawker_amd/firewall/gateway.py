@@ -931,6 +931,11 @@ class GatewayManager:
         b.settimeout(600)
 
         def pump(src, dst):
+            try:
+                threading.current_thread().name = (
+                    f"pump-{src.fileno()}to{dst.fileno()}")
+            except OSError:
+                pass
             # 256 KiB zero-copy-ish relay: recv_into a reused buffer
             # halves allocator churn vs recv() at bulk rates
             buf = bytearray(262144)
