@@ -3,6 +3,7 @@
 ensureRunning + controlplane/adminclient Dial)."""
 from __future__ import annotations
 
+import errno
 import os
 import socket
 import subprocess
@@ -87,10 +88,27 @@ class CPClient:
 
     # -- requests ---------------------------------------------------------------
     def _connect(self) -> socket.socket:
-        s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
-        s.settimeout(self.timeout)
-        s.connect(str(admin_sock_path()))
-        return s
+        # unix connect() returns EAGAIN instantly when the daemon's accept
+        # queue is momentarily full (fleet cold-start storms) and
+        # ECONNREFUSED in the bind→listen window of a daemon that is
+        # coming up; both are transient — retry briefly before declaring
+        # the control plane unreachable.
+        deadline = time.monotonic() + min(self.timeout, 5.0)
+        delay = 0.005
+        while True:
+            s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+            s.settimeout(self.timeout)
+            try:
+                s.connect(str(admin_sock_path()))
+                return s
+            except OSError as e:
+                s.close()
+                if (e.errno not in (errno.EAGAIN, errno.ECONNREFUSED,
+                                    errno.ECONNRESET)
+                        or time.monotonic() >= deadline):
+                    raise
+                time.sleep(delay)
+                delay = min(delay * 2, 0.1)
 
     def _ping(self) -> bool:
         try:

@@ -591,7 +591,11 @@ class CPDaemon:
             sock_path.unlink(missing_ok=True)
             listener.bind(str(sock_path))
             os.chmod(sock_path, 0o600)
-            listener.listen(16)
+            # unix SOCK_STREAM connect() fails with EAGAIN the moment the
+            # accept queue is full (no TCP-style SYN retry), so size the
+            # backlog for a whole fleet cold-starting at once: 32 agents ×
+            # several CP round-trips each showed EAGAIN bursts at 16
+            listener.listen(512)
         except OSError as e:
             log.error("admin_listen_failed", err=str(e))
             return 1

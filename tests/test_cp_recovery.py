@@ -118,3 +118,69 @@ def test_cp_kill_fails_closed_then_recovers(isolated_env, tmp_path,
             pass
         CPClient(auto_start=False).stop()
         orch.close()
+
+
+def test_client_connect_rides_out_backlog_storm(isolated_env, monkeypatch,
+                                                tmp_path):
+    """Fleet cold-start storms transiently fill cpd's accept queue; unix
+    connect() then fails with EAGAIN immediately (no TCP SYN retry).
+    CPClient._connect must retry past the burst instead of reporting
+    'control plane unreachable' (r02: 207/16000 loop failures at 32-way
+    before the fix)."""
+    import socket
+    import threading
+    import time as _t
+
+    from clawker_amd.controlplane import client as cpc
+    from clawker_amd.engine import wire
+
+    sock = tmp_path / "cp.sock"
+    monkeypatch.setattr(cpc, "admin_sock_path", lambda: sock)
+
+    lst = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    lst.bind(str(sock))
+    lst.listen(0)        # minimal backlog → trivially saturated
+
+    # saturate the queue: park connects until the backlog refuses more
+    parked = []
+    for _ in range(16):
+        s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        s.setblocking(False)
+        try:
+            s.connect(str(sock))
+            parked.append(s)
+        except BlockingIOError:
+            s.close()
+            break
+    else:
+        lst.close()
+        raise AssertionError("could not saturate backlog")
+
+    def drain_later():
+        _t.sleep(0.2)    # hold the storm long enough to defeat one try
+        while True:
+            try:
+                lst.settimeout(0.5)
+                c, _ = lst.accept()
+            except socket.timeout:
+                return
+            try:
+                c.settimeout(0.3)    # parked conns never speak — skip them
+                req = wire.recv_frame(c)
+                if req:
+                    wire.send_frame(c, {"ok": True, "echo": req.get("op")})
+            except OSError:
+                pass
+            c.close()
+
+    t = threading.Thread(target=drain_later, daemon=True)
+    t.start()
+    try:
+        cl = cpc.CPClient(auto_start=False)
+        resp = cl.request({"op": "ping"})
+        assert resp["ok"] and resp["echo"] == "ping"
+    finally:
+        for s in parked:
+            s.close()
+        t.join(timeout=5)
+        lst.close()
