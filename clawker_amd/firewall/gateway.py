@@ -278,7 +278,7 @@ class GatewayManager:
                 tmp.unlink(missing_ok=True)
                 lst = bind_unix(tmp)
                 os.chmod(tmp, 0o666)    # in-sandbox ckgw connects as root-inside
-                lst.listen(64)
+                lst.listen(256)
                 tmp.rename(path)
                 lst.settimeout(0.5)
                 gw.listeners.append(lst)

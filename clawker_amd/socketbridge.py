@@ -82,7 +82,7 @@ class SocketBridgeManager:
                     log.warn("bridge_bind_failed", sock=sock_name, err=str(e))
                     continue
                 os.chmod(tmp, 0o666)
-                lst.listen(16)
+                lst.listen(128)
                 path.unlink(missing_ok=True)
                 tmp.rename(path)
                 lst.settimeout(0.5)

@@ -55,7 +55,7 @@ int tcp_listen(const char* ip, int port) {
   inet_pton(AF_INET, ip, &addr.sin_addr);
   if (bind(fd, reinterpret_cast<sockaddr*>(&addr), sizeof addr) < 0)
     die("bind %s:%d", ip, port);
-  if (listen(fd, 64) < 0) die("listen");
+  if (listen(fd, 256) < 0) die("listen");
   fcntl(fd, F_SETFL, O_NONBLOCK);   // accept-drain loop must not block
   return fd;
 }

@@ -130,7 +130,7 @@ inline std::vector<uint8_t> b64_decode(const std::string& in) {
 }
 
 // -------------------------------------------------------------- unix sock --
-inline int unix_listen(const std::string& path, int backlog = 16) {
+inline int unix_listen(const std::string& path, int backlog = 128) {
   int fd = socket(AF_UNIX, SOCK_STREAM | SOCK_CLOEXEC, 0);
   if (fd < 0) return -1;
   sockaddr_un addr{};
