@@ -91,9 +91,14 @@ class CPClient:
         # unix connect() returns EAGAIN instantly when the daemon's accept
         # queue is momentarily full (fleet cold-start storms) and
         # ECONNREFUSED in the bind→listen window of a daemon that is
-        # coming up; both are transient — retry briefly before declaring
-        # the control plane unreachable.
-        deadline = time.monotonic() + min(self.timeout, 5.0)
+        # coming up. Both are transient, but with very different budgets:
+        # a saturated-but-alive daemon drains within seconds, while
+        # ECONNREFUSED against a STALE socket file (daemon SIGKILLed) is
+        # persistent — retrying it long makes every liveness ping of a
+        # dead daemon hang, which stalled CP crash-recovery under load.
+        start = time.monotonic()
+        deadline_eagain = start + min(self.timeout, 5.0)
+        deadline_refused = start + 0.3   # bind→listen gap is microseconds
         delay = 0.005
         while True:
             s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
@@ -103,9 +108,13 @@ class CPClient:
                 return s
             except OSError as e:
                 s.close()
-                if (e.errno not in (errno.EAGAIN, errno.ECONNREFUSED,
-                                    errno.ECONNRESET)
-                        or time.monotonic() >= deadline):
+                now = time.monotonic()
+                if e.errno == errno.EAGAIN and now < deadline_eagain:
+                    pass
+                elif (e.errno in (errno.ECONNREFUSED, errno.ECONNRESET)
+                      and now < deadline_refused):
+                    pass
+                else:
                     raise
                 time.sleep(delay)
                 delay = min(delay * 2, 0.1)
